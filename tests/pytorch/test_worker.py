@@ -1,0 +1,54 @@
+"""PyTorch worker-task unit tests (reference tests/pytorch/tasks/test_worker.py)."""
+
+import os
+from unittest import mock
+
+import pytest
+import torch
+
+from tf_yarn_amd.pytorch.experiment import DataLoaderArgs
+from tf_yarn_amd.pytorch.tasks import worker
+
+
+def test_get_device_round_robin(monkeypatch):
+    monkeypatch.setenv("MIYARN_GPU_IDS", "4,5")
+    gpu_ids = worker._get_gpu_ids()
+    assert gpu_ids == [4, 5]
+    with mock.patch.object(torch.cuda, "is_available", return_value=True):
+        assert worker._get_device(gpu_ids, 0) == "cuda:4"
+        assert worker._get_device(gpu_ids, 1) == "cuda:5"
+        assert worker._get_device(gpu_ids, 2) == "cuda:4"
+    assert worker._get_device([], 0) == "cpu"
+
+
+def test_get_device_cpu_when_no_cuda():
+    with mock.patch.object(torch.cuda, "is_available", return_value=False):
+        assert worker._get_device([3], 0) == "cpu"
+
+
+def test_backend_selection():
+    assert worker._get_collective_ops_backend("cuda:0") == "nccl"
+    assert worker._get_collective_ops_backend("cpu") == "gloo"
+
+
+def test_create_dataloader_map_dataset_uses_sampler():
+    ds = torch.utils.data.TensorDataset(torch.arange(100).float())
+    loader = worker._create_dataloader(
+        ds, DataLoaderArgs(batch_size=10, pin_memory=False),
+        rank=0, world_size=2)
+    assert isinstance(loader.sampler,
+                      torch.utils.data.distributed.DistributedSampler)
+    batches = list(loader)
+    assert len(batches) == 5  # 100 / 2 ranks / batch 10
+
+
+def test_create_dataloader_iterable_passthrough():
+    class It(torch.utils.data.IterableDataset):
+        def __iter__(self):
+            return iter(torch.arange(20).float().split(1))
+
+    loader = worker._create_dataloader(
+        It(), DataLoaderArgs(batch_size=5, pin_memory=False),
+        rank=0, world_size=2)
+    assert loader.sampler is None or not isinstance(
+        loader.sampler, torch.utils.data.distributed.DistributedSampler)

@@ -1,0 +1,111 @@
+"""End-to-end: run_on_yarn (pytorch flavor) on CPU with chief + worker.
+
+Exercises the full stack: spawner -> worker task -> KV master election ->
+gloo process group via the KV rendezvous store -> BucketedDataParallel ->
+user main_fn -> checkpoint -> event aggregation -> Metrics.
+"""
+
+import os
+import sys
+
+import cloudpickle
+import pytest
+import torch
+from torch import nn
+
+from tf_yarn_amd import TaskSpec
+
+# Task processes cannot import this test module: pickle its closures by value.
+cloudpickle.register_pickle_by_value(sys.modules[__name__])
+from tf_yarn_amd.pytorch import (DataLoaderArgs, PytorchExperiment,
+                                 run_on_yarn)
+
+
+def _experiment_fn(model_dir):
+    import torch
+    from torch import nn
+
+    from tf_yarn_amd.pytorch import DataLoaderArgs, PytorchExperiment
+    from tf_yarn_amd.pytorch import model_ckpt
+
+    def main_fn(model, loader, device, rank, tb_writer):
+        import torch.distributed as dist
+        opt = torch.optim.SGD(model.parameters(), lr=0.05)
+        losses = []
+        for epoch in range(2):
+            for x, y in loader:
+                opt.zero_grad()
+                loss = nn.functional.mse_loss(model(x), y)
+                loss.backward()
+                opt.step()
+                losses.append(loss.item())
+        dist.barrier()
+        if rank == 0:
+            model_ckpt.save_ckpt(model_dir, model, opt, epoch=2,
+                                 final_loss=losses[-1])
+        if tb_writer is not None:
+            tb_writer.add_scalar("loss", losses[-1], 2)
+
+    torch.manual_seed(0)
+    model = nn.Sequential(nn.Linear(8, 16), nn.ReLU(), nn.Linear(16, 1))
+    x = torch.randn(64, 8)
+    y = x.sum(dim=1, keepdim=True)
+    dataset = torch.utils.data.TensorDataset(x, y)
+    return PytorchExperiment(
+        model=model,
+        main_fn=main_fn,
+        train_dataset=dataset,
+        dataloader_args=DataLoaderArgs(batch_size=8, pin_memory=False),
+        tensorboard_hdfs_dir=os.path.join(model_dir, "tb"),
+    )
+
+
+@pytest.mark.timeout(180)
+def test_run_on_yarn_pytorch_cpu(tmp_path):
+    model_dir = str(tmp_path / "model")
+    from functools import partial
+    metrics = run_on_yarn(
+        partial(_experiment_fn, model_dir),
+        {
+            "chief": TaskSpec(memory=512, vcores=1),
+            "worker": TaskSpec(memory=512, vcores=1, instances=1),
+        },
+        base_dir=str(tmp_path / "app"),
+    )
+    assert metrics is not None
+    assert metrics.total_training_duration is not None
+    assert metrics.total_training_duration > 0
+    # rank 0 (chief) wrote the checkpoint
+    assert os.path.exists(os.path.join(model_dir, "model_2.pt"))
+    # per-worker tensorboard logs were uploaded for both ranks
+    tb_dir = os.path.join(model_dir, "tb")
+    assert os.path.isdir(os.path.join(tb_dir, "worker_0"))
+    assert os.path.isdir(os.path.join(tb_dir, "worker_1"))
+
+
+@pytest.mark.timeout(120)
+def test_run_on_yarn_failure_propagates(tmp_path):
+    from tf_yarn_amd import RunFailed
+
+    def bad_experiment():
+        from tf_yarn_amd.pytorch import DataLoaderArgs, PytorchExperiment
+        import torch
+        from torch import nn
+
+        def main_fn(model, loader, device, rank, tb_writer):
+            raise RuntimeError("deliberate failure for test")
+
+        return PytorchExperiment(
+            model=nn.Linear(2, 2),
+            main_fn=main_fn,
+            train_dataset=torch.utils.data.TensorDataset(
+                torch.randn(8, 2), torch.randn(8, 2)),
+            dataloader_args=DataLoaderArgs(batch_size=4, pin_memory=False),
+        )
+
+    with pytest.raises(RunFailed):
+        run_on_yarn(
+            bad_experiment,
+            {"chief": TaskSpec(memory=512, vcores=1)},
+            base_dir=str(tmp_path / "app"),
+        )

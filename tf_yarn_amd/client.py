@@ -141,6 +141,14 @@ def _task_env(task: ContainerTask,
     """Build the per-container environment contract
     (reference ``client.py:108-133`` + TB_* plumbing ``client.py:213-219``)."""
     child = dict(os.environ)
+    # Ship the framework itself into the task env (the reference zips
+    # tf_yarn into each container, client.py:108-133): here it is enough to
+    # put the package's parent directory on the child's PYTHONPATH.
+    pkg_root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    existing_pp = child.get("PYTHONPATH", "")
+    if pkg_root not in existing_pp.split(os.pathsep):
+        child["PYTHONPATH"] = (pkg_root + os.pathsep + existing_pp
+                               if existing_pp else pkg_root)
     child[constants.ENV_CONTAINER_ID] = f"{task.type}_{task.id}"
     child[constants.ENV_KV_ADDR] = kv_addr
     child[constants.ENV_APP_ID] = app_id
@@ -204,9 +212,12 @@ def _aggregate_events(client: KVClient,
     """Event-listener thread body (reference ``client.py:633-657``):
     consume the KV watch stream and bucket events per task."""
     for key, value in client.events(""):
+        if key.startswith("c10d/"):
+            continue  # binary rendezvous traffic, not lifecycle events
         if "/" in key:
             task, stage = key.split("/", 1)
-            events.setdefault(task, {})[stage] = value.decode()
+            events.setdefault(task, {})[stage] = value.decode(
+                errors="replace")
 
 
 def _setup_cluster(task_specs: TaskSpecs,

@@ -110,6 +110,36 @@ class KVServer:
                             _send_frame(conn, ("ok", self._data[key]))
                         else:
                             _send_frame(conn, ("timeout", None))
+                elif op == "add":
+                    # Atomic counter (c10d Store.add semantics): value is
+                    # stored as its decimal-string bytes.
+                    key, amount = req[1], req[2]
+                    with self._cond:
+                        current = int(self._data.get(key, b"0") or b"0")
+                        current += amount
+                        self._data[key] = str(current).encode()
+                        self._cond.notify_all()
+                    self._notify_watchers(key, str(current).encode())
+                    _send_frame(conn, ("ok", current))
+                elif op == "cas":
+                    # compare_set (c10d semantics): set to desired iff the
+                    # current value equals expected, or the key is absent
+                    # and expected is empty.  Returns the resulting value.
+                    key, expected, desired = req[1], req[2], req[3]
+                    with self._cond:
+                        cur = self._data.get(key)
+                        if (cur == expected
+                                or (cur is None and expected == b"")):
+                            self._data[key] = desired
+                            self._cond.notify_all()
+                            result = desired
+                            changed = True
+                        else:
+                            result = cur if cur is not None else expected
+                            changed = False
+                    if changed:
+                        self._notify_watchers(key, desired)
+                    _send_frame(conn, ("ok", result))
                 elif op == "del":
                     with self._cond:
                         self._data.pop(req[1], None)
@@ -152,6 +182,10 @@ class KVServer:
         with self._cond:
             self._data[key] = value
             self._cond.notify_all()
+        self._notify_watchers(key, value)
+
+    def _notify_watchers(self, key: str, value: bytes) -> None:
+        with self._cond:
             watchers = list(self._watchers)
         dead = []
         for prefix, wsock, wlock in watchers:
@@ -237,6 +271,16 @@ class KVClient:
         if status == "timeout":
             raise TimeoutError(f"kv wait timed out after {timeout}s: {key!r}")
         return value
+
+    def add(self, key: str, amount: int) -> int:
+        """Atomically add to a decimal counter; returns the new value."""
+        return self._call("add", key, amount)
+
+    def compare_set(self, key: str, expected: bytes,
+                    desired: bytes) -> bytes:
+        """Set *key* to *desired* iff its value equals *expected* (or the
+        key is absent and expected is empty); returns the resulting value."""
+        return self._call("cas", key, bytes(expected), bytes(desired))
 
     def delete(self, key: str) -> None:
         self._call("del", key)

@@ -1,0 +1,8 @@
+from tf_yarn_amd.pytorch.client import run_on_yarn
+from tf_yarn_amd.pytorch.experiment import (DataLoaderArgs,
+                                            DistributedDataParallelArgs,
+                                            PytorchExperiment)
+from tf_yarn_amd.topologies import NodeLabel, TaskSpec
+
+__all__ = ["run_on_yarn", "PytorchExperiment", "DataLoaderArgs",
+           "DistributedDataParallelArgs", "TaskSpec", "NodeLabel"]
