@@ -1,0 +1,168 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: Criteo wide-and-deep examples/sec on MI355X.
+
+Contract (driver): ``python bench.py --gpus N --steps K --warmup W``.
+For N>1 the driver launches it under ``torch.distributed.run`` with one
+rank per GPU (RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* in env), RCCL over xGMI.
+W untimed warmup steps, then exactly K timed steps bracketed by
+barrier + torch.cuda.synchronize on both sides; MAX time over ranks;
+rank 0 prints ONE JSON line.
+
+Metric (BASELINE.json): examples/sec (whole node), Criteo wide-and-deep,
+synthetic Criteo-shaped data, random-init weights, bf16 compute, weak
+scaling (per-GPU batch fixed).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+import torch.distributed as dist
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from tf_yarn_amd.models.synthetic import synthetic_criteo_batch  # noqa: E402
+from tf_yarn_amd.models.wide_deep import WideAndDeep  # noqa: E402
+from tf_yarn_amd.ops.optim import FusedSGD  # noqa: E402
+from tf_yarn_amd.parallel.ddp import BucketedDataParallel  # noqa: E402
+
+PER_GPU_BATCH = 16384
+TABLE_ROWS_PER_FEATURE = 1_000_000
+EMBEDDING_DIM = 16
+HIDDEN = (1024, 512, 256)
+N_DATA_BATCHES = 8  # distinct synthetic batches cycled through
+
+
+def log(msg: str) -> None:
+    print(f"[bench] {msg}", file=sys.stderr, flush=True)
+
+
+def main() -> None:
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=20)
+    parser.add_argument("--warmup", type=int, default=5)
+    parser.add_argument("--batch", type=int, default=PER_GPU_BATCH)
+    parser.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    parser.add_argument("--table-rows", type=int,
+                        default=TABLE_ROWS_PER_FEATURE,
+                        help="rows per categorical table (tests use small)")
+    args = parser.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world_size = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+
+    use_gpu = torch.cuda.is_available()
+    device = f"cuda:{local_rank}" if use_gpu else "cpu"
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+    backend = "nccl" if use_gpu else "gloo"
+
+    if world_size > 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29500")
+        dist.init_process_group(backend, rank=rank, world_size=world_size)
+
+    compute_dtype = (torch.bfloat16 if args.dtype == "bf16" and use_gpu
+                     else torch.float32)
+    table_sizes = [args.table_rows] * 26
+
+    torch.manual_seed(1234)  # identical random-init across ranks
+    model = WideAndDeep(table_sizes=table_sizes,
+                        embedding_dim=EMBEDDING_DIM, hidden=HIDDEN,
+                        compute_dtype=compute_dtype).to(device)
+    ddp = BucketedDataParallel(model) if world_size > 1 else model
+    module = ddp.module if world_size > 1 else model
+
+    lr = 0.02
+    opt = FusedSGD([p for p in model.parameters()
+                    if not getattr(p, "_miyarn_sparse", False)], lr=lr)
+    loss_fn = torch.nn.BCEWithLogitsLoss()
+
+    # Pre-generate distinct synthetic batches on-device (data=synthetic;
+    # varying ids each step so the gather/scatter path is exercised).
+    batches = []
+    for i in range(N_DATA_BATCHES):
+        dense, ids, labels = synthetic_criteo_batch(
+            args.batch, table_sizes, device=device, seed=1000 * rank + i)
+        batches.append((dense, ids, labels))
+
+    def step(i: int) -> float:
+        dense, ids, labels = batches[i % N_DATA_BATCHES]
+        opt.zero_grad(set_to_none=False)
+        logits = ddp(dense, ids)
+        loss = loss_fn(logits.float(), labels)
+        loss.backward()
+        opt.step()
+        module.apply_sparse_updates(lr)
+        return loss
+
+    log(f"rank {rank}/{world_size} device={device} "
+        f"dtype={compute_dtype} batch/gpu={args.batch}")
+    for i in range(args.warmup):
+        step(i)
+    if use_gpu:
+        torch.cuda.synchronize()
+    if world_size > 1:
+        dist.barrier()
+    t0 = time.perf_counter()
+    last_loss = None
+    for i in range(args.steps):
+        last_loss = step(args.warmup + i)
+    if use_gpu:
+        torch.cuda.synchronize()
+    if world_size > 1:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if backend == "nccl" else "cpu")
+    if world_size > 1:
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    if rank == 0:
+        ms_per_step = elapsed / args.steps * 1000.0
+        examples_per_sec = args.batch * world_size * args.steps / elapsed
+        result = {
+            "metric": "examples/sec (whole node), Criteo wide-and-deep",
+            "value": examples_per_sec,
+            "unit": "examples/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": args.dtype if use_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "wide_and_deep_criteo",
+                "global_batch": args.batch * world_size,
+                "per_gpu_batch": args.batch,
+                "dense_features": 13,
+                "sparse_features": 26,
+                "embedding_dim": EMBEDDING_DIM,
+                "table_rows_per_feature": args.table_rows,
+                "hidden": list(HIDDEN),
+                "parallelism": f"dp{world_size}",
+                "final_loss": float(last_loss.item())
+                if last_loss is not None else None,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if world_size > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,40 @@
+"""Build the MI355X (gfx950) HIP kernel extension in-tree.
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+produces ``tf_yarn_amd/ops/_C.*.so`` next to its Python wrappers so the
+built library travels with the source tree.
+"""
+
+import os
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from setuptools import find_packages, setup  # noqa: E402
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+CSRC = os.path.join("tf_yarn_amd", "ops", "csrc")
+
+ext = CUDAExtension(
+    name="tf_yarn_amd.ops._C",
+    sources=[
+        os.path.join(CSRC, "bindings.cpp"),
+        os.path.join(CSRC, "fused_optimizers.hip"),
+        os.path.join(CSRC, "embedding.hip"),
+        os.path.join(CSRC, "elementwise.hip"),
+    ],
+    extra_compile_args={
+        "cxx": ["-O3"],
+        "nvcc": ["-O3", "--offload-arch=gfx950"],
+    },
+)
+
+setup(
+    name="tf_yarn_amd",
+    version="0.1.0",
+    description=("MI355X-native distributed-training launcher with "
+                 "criteo/tf-yarn's capabilities"),
+    packages=find_packages(include=["tf_yarn_amd", "tf_yarn_amd.*"]),
+    ext_modules=[ext],
+    cmdclass={"build_ext": BuildExtension},
+)

@@ -1,0 +1,102 @@
+"""CPU-side tests for the op wrappers: the CPU reference implementations
+must match stock torch optimizers/ops exactly (they are the baseline the
+GPU numerics tests compare against)."""
+
+import pytest
+import torch
+from torch import nn
+
+from tf_yarn_amd import ops
+from tf_yarn_amd.ops.optim import (FusedAdadelta, FusedAdagrad, FusedAdam,
+                                   FusedSGD)
+
+
+def _clone_params(model):
+    return [p.detach().clone() for p in model.parameters()]
+
+
+def _train(model, opt, steps=5, seed=7):
+    torch.manual_seed(seed)
+    xs = [torch.randn(8, 6) for _ in range(steps)]
+    ys = [torch.randn(8, 3) for _ in range(steps)]
+    for x, y in zip(xs, ys):
+        opt.zero_grad()
+        nn.functional.mse_loss(model(x), y).backward()
+        opt.step()
+
+
+@pytest.mark.parametrize("fused_cls,torch_cls,kwargs", [
+    (FusedSGD, torch.optim.SGD, dict(lr=0.1)),
+    (FusedSGD, torch.optim.SGD,
+     dict(lr=0.1, momentum=0.9, weight_decay=0.01)),
+    (FusedSGD, torch.optim.SGD,
+     dict(lr=0.1, momentum=0.9, nesterov=True)),
+    (FusedAdam, torch.optim.Adam,
+     dict(lr=0.01, weight_decay=0.001)),
+    (FusedAdagrad, torch.optim.Adagrad, dict(lr=0.05)),
+    (FusedAdadelta, torch.optim.Adadelta, dict(lr=0.9)),
+])
+def test_fused_optimizer_matches_torch(fused_cls, torch_cls, kwargs):
+    torch.manual_seed(0)
+    m1 = nn.Sequential(nn.Linear(6, 16), nn.Tanh(), nn.Linear(16, 3))
+    torch.manual_seed(0)
+    m2 = nn.Sequential(nn.Linear(6, 16), nn.Tanh(), nn.Linear(16, 3))
+    _train(m1, fused_cls(m1.parameters(), **kwargs))
+    _train(m2, torch_cls(m2.parameters(), **kwargs))
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-5), \
+            f"{fused_cls.__name__} diverges from {torch_cls.__name__}"
+
+
+def test_fused_adamw_matches_torch():
+    torch.manual_seed(0)
+    m1 = nn.Linear(6, 3)
+    torch.manual_seed(0)
+    m2 = nn.Linear(6, 3)
+    _train(m1, FusedAdam(m1.parameters(), lr=0.01, weight_decay=0.05,
+                         adamw=True))
+    _train(m2, torch.optim.AdamW(m2.parameters(), lr=0.01,
+                                 weight_decay=0.05))
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-5)
+
+
+def test_emb_fwd_cpu_matches_index_select():
+    table = torch.randn(50, 8)
+    ids = torch.randint(0, 50, (30,))
+    out = ops.emb_fwd(table, ids)
+    assert torch.equal(out, table.index_select(0, ids))
+    out16 = ops.emb_fwd(table, ids, out_bf16=True)
+    assert out16.dtype == torch.bfloat16
+
+
+def test_emb_bwd_sgd_cpu():
+    table = torch.randn(20, 4)
+    table0 = table.clone()
+    ids = torch.tensor([3, 3, 7])
+    grad = torch.randn(3, 4)
+    ops.emb_bwd_sgd(table, ids, grad, lr=0.5, scale=2.0)
+    expected = table0.index_add(0, ids, grad, alpha=-1.0)
+    assert torch.allclose(table, expected, atol=1e-6)
+
+
+def test_bias_relu_autograd_matches_torch():
+    x = torch.randn(10, 6, requires_grad=True)
+    b = torch.randn(6, requires_grad=True)
+    y = ops.bias_relu(x, b)
+    loss = (y ** 2).sum()
+    loss.backward()
+    x2 = x.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    y2 = torch.relu(x2 + b2)
+    ((y2 ** 2).sum()).backward()
+    assert torch.allclose(y, y2)
+    assert torch.allclose(x.grad, x2.grad)
+    assert torch.allclose(b.grad, b2.grad)
+
+
+def test_convert_scaled_cpu():
+    src = torch.randn(13)
+    dst = torch.empty(13, dtype=torch.bfloat16)
+    ops.convert_scaled(src, dst, 2.0)
+    assert torch.allclose(dst.float(), (src * 2).to(torch.bfloat16).float())

@@ -1,0 +1,192 @@
+"""Criteo wide-and-deep model, MI355X-first.
+
+The reference's target workload family (CTR models trained through the
+Keras/Estimator paths; SURVEY §5 "target workloads are CTR/tabular").
+Design decisions for MI355X:
+
+* All 26 categorical tables are fused into ONE [total_rows, D] fp32 buffer;
+  Python pre-adds per-feature row offsets so the gather/scatter kernels see
+  flat ids (:mod:`tf_yarn_amd.ops` ``emb_fwd`` / ``emb_bwd_sgd``).
+* The embedding gradient never materializes as a dense table: backward
+  stashes (ids, grad_rows); at step time the rows are allgathered across
+  data-parallel ranks (equal batch sizes => equal counts, so
+  ``all_gather_into_tensor`` over RCCL) and applied with the fused
+  atomic scatter+SGD kernel scaled by 1/world.  This replaces the dense
+  allreduce that would move the whole table every step.
+* Dense compute runs bf16 (``emb_fwd`` fuses the fp32->bf16 cast into the
+  gather); dense params keep fp32 master weights.
+* The MLP epilogue uses the fused bias+ReLU kernel.
+"""
+
+from __future__ import annotations
+
+import logging
+import math
+from typing import List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+from torch import nn
+
+from tf_yarn_amd import ops
+
+logger = logging.getLogger(__name__)
+
+# Criteo display-advertising schema: 13 integer (dense) + 26 categorical.
+CRITEO_DENSE = 13
+CRITEO_SPARSE = 26
+# Per-feature hash sizes for the synthetic Criteo-1TB-shaped benchmark
+# (hashed to 1M rows max per feature, a common Criteo preprocessing).
+DEFAULT_TABLE_SIZES = [1_000_000] * CRITEO_SPARSE
+
+
+class _FusedLookup(torch.autograd.Function):
+    """Gather via the HIP kernel; backward stashes sparse (ids, grad)."""
+
+    @staticmethod
+    def forward(ctx, table, flat_ids, out_bf16, sink):
+        out = ops.emb_fwd(table, flat_ids, out_bf16)
+        ctx.sink = sink
+        ctx.save_for_backward(flat_ids)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        (flat_ids,) = ctx.saved_tensors
+        ctx.sink.append((flat_ids, grad_out.contiguous()))
+        return None, None, None, None
+
+
+class SparseEmbedding(nn.Module):
+    """Fused multi-table embedding with sparse-allgather DP sync.
+
+    ``forward(ids [B, F])`` -> ``[B, F*dim]``.
+    After ``loss.backward()`` call :meth:`apply_sparse_updates` (the
+    training loop or the framework optimizer wrapper does this).
+    """
+
+    def __init__(self, table_sizes: List[int], dim: int,
+                 out_bf16: bool = False, init_std: Optional[float] = None):
+        super().__init__()
+        self.table_sizes = list(table_sizes)
+        self.dim = dim
+        self.out_bf16 = out_bf16
+        total = sum(table_sizes)
+        offsets = torch.tensor(
+            [0] + list(torch.cumsum(torch.tensor(table_sizes), 0)[:-1]),
+            dtype=torch.int64)
+        self.register_buffer("offsets", offsets, persistent=True)
+        std = init_std if init_std is not None else 1.0 / math.sqrt(dim)
+        weight = torch.randn(total, dim) * std
+        self.weight = nn.Parameter(weight)
+        self.weight._miyarn_sparse = True  # dense reducer must skip it
+        self._sink: List[Tuple[torch.Tensor, torch.Tensor]] = []
+
+    def forward(self, ids: torch.Tensor) -> torch.Tensor:
+        b, f = ids.shape
+        assert f == len(self.table_sizes), "feature count mismatch"
+        flat = (ids + self.offsets.unsqueeze(0)).reshape(-1)
+        out = _FusedLookup.apply(self.weight, flat, self.out_bf16,
+                                 self._sink)
+        return out.reshape(b, f * self.dim)
+
+    def pending_grads(self) -> List[Tuple[torch.Tensor, torch.Tensor]]:
+        return self._sink
+
+    @torch.no_grad()
+    def apply_sparse_updates(self, lr: float,
+                             process_group=None) -> None:
+        """Allgather sparse grads across DP ranks and apply the fused
+        scatter+SGD update, scaled by 1/world_size."""
+        if not self._sink:
+            return
+        world = (dist.get_world_size(process_group)
+                 if dist.is_available() and dist.is_initialized() else 1)
+        for flat_ids, grad in self._sink:
+            grad2d = grad.reshape(flat_ids.numel(), self.dim)
+            if world > 1:
+                n = flat_ids.numel()
+                all_ids = torch.empty(n * world, dtype=flat_ids.dtype,
+                                      device=flat_ids.device)
+                all_grads = torch.empty((n * world, self.dim),
+                                        dtype=grad2d.dtype,
+                                        device=grad2d.device)
+                dist.all_gather_into_tensor(all_ids, flat_ids,
+                                            group=process_group)
+                dist.all_gather_into_tensor(all_grads, grad2d,
+                                            group=process_group)
+            else:
+                all_ids, all_grads = flat_ids, grad2d
+            ops.emb_bwd_sgd(self.weight.data, all_ids, all_grads,
+                            lr=lr, scale=1.0 / world)
+        self._sink.clear()
+
+    def clear_pending(self) -> None:
+        self._sink.clear()
+
+
+class FusedLinearReLU(nn.Module):
+    """Linear (library GEMM) + fused bias+ReLU epilogue kernel."""
+
+    def __init__(self, in_features: int, out_features: int,
+                 dtype=torch.float32):
+        super().__init__()
+        self.weight = nn.Parameter(
+            torch.empty(out_features, in_features, dtype=dtype))
+        self.bias = nn.Parameter(torch.zeros(out_features, dtype=dtype))
+        nn.init.kaiming_uniform_(self.weight, a=math.sqrt(5))
+
+    def forward(self, x):
+        y = x.matmul(self.weight.t())
+        return ops.bias_relu(y, self.bias)
+
+
+class WideAndDeep(nn.Module):
+    """Wide (per-id scalar weights) + deep (embeddings -> MLP) CTR model."""
+
+    def __init__(self,
+                 dense_dim: int = CRITEO_DENSE,
+                 table_sizes: Optional[List[int]] = None,
+                 embedding_dim: int = 16,
+                 hidden: Tuple[int, ...] = (1024, 512, 256),
+                 compute_dtype: torch.dtype = torch.float32):
+        super().__init__()
+        table_sizes = table_sizes or DEFAULT_TABLE_SIZES
+        self.compute_dtype = compute_dtype
+        bf16 = compute_dtype == torch.bfloat16
+        self.deep_embedding = SparseEmbedding(table_sizes, embedding_dim,
+                                              out_bf16=bf16)
+        # Wide part: one scalar weight per categorical id (linear-in-one-hot)
+        self.wide_embedding = SparseEmbedding(table_sizes, 1, out_bf16=bf16,
+                                              init_std=0.01)
+        self.wide_dense = nn.Linear(dense_dim, 1)
+        layers: List[nn.Module] = []
+        in_dim = dense_dim + len(table_sizes) * embedding_dim
+        for h in hidden:
+            layers.append(FusedLinearReLU(in_dim, h))
+            in_dim = h
+        self.mlp = nn.Sequential(*layers)
+        self.head = nn.Linear(in_dim, 1)
+        if bf16:
+            # dense compute in bf16; masters are managed by the optimizer
+            self.wide_dense = self.wide_dense.to(torch.bfloat16)
+            self.mlp = self.mlp.to(torch.bfloat16)
+            self.head = self.head.to(torch.bfloat16)
+
+    def forward(self, dense: torch.Tensor,
+                sparse_ids: torch.Tensor) -> torch.Tensor:
+        dense = dense.to(self.compute_dtype)
+        deep_in = torch.cat(
+            [dense, self.deep_embedding(sparse_ids)], dim=1)
+        deep_out = self.head(self.mlp(deep_in))
+        wide_out = (self.wide_embedding(sparse_ids).sum(dim=1, keepdim=True)
+                    + self.wide_dense(dense))
+        return (deep_out + wide_out).squeeze(1)
+
+    def apply_sparse_updates(self, lr: float, process_group=None) -> None:
+        self.deep_embedding.apply_sparse_updates(lr, process_group)
+        self.wide_embedding.apply_sparse_updates(lr, process_group)
+
+    def clear_pending(self) -> None:
+        self.deep_embedding.clear_pending()
+        self.wide_embedding.clear_pending()

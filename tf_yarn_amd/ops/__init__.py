@@ -1,0 +1,208 @@
+"""MI355X HIP kernel library: Python wrappers + CPU references.
+
+Policy (required for the GPU-native check): on a machine with a GPU the
+HIP extension MUST be present — ops raise immediately if it failed to
+import, instead of silently falling back to eager torch.  On CPU-only
+machines (the CI container) the same functions run reference torch
+implementations so every numerics test has a CPU baseline.
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import Optional
+
+import torch
+
+logger = logging.getLogger(__name__)
+
+try:
+    from tf_yarn_amd.ops import _C  # built by setup.py build_ext --inplace
+    HAVE_EXT = True
+except ImportError as _e:  # pragma: no cover - exercised only sans build
+    _C = None
+    HAVE_EXT = False
+    _IMPORT_ERROR = _e
+
+
+def _require_ext() -> None:
+    if not HAVE_EXT:
+        raise RuntimeError(
+            "tf_yarn_amd.ops._C is not built but a GPU tensor was passed. "
+            "Run `python setup.py build_ext --inplace` "
+            f"(import error: {_IMPORT_ERROR})")
+
+
+def _on_gpu(*tensors: torch.Tensor) -> bool:
+    return any(t.is_cuda for t in tensors if isinstance(t, torch.Tensor))
+
+
+# ---- fused optimizers ------------------------------------------------------
+
+def fused_sgd(param: torch.Tensor, grad: torch.Tensor,
+              momentum_buf: Optional[torch.Tensor] = None,
+              param_bf16: Optional[torch.Tensor] = None,
+              *, lr: float, momentum: float = 0.0, dampening: float = 0.0,
+              weight_decay: float = 0.0, nesterov: bool = False,
+              first_step: bool = False, grad_scale: float = 1.0) -> None:
+    if _on_gpu(param, grad):
+        _require_ext()
+        _C.fused_sgd(param, grad, momentum_buf, param_bf16, lr, momentum,
+                     dampening, weight_decay, nesterov, first_step,
+                     grad_scale)
+        return
+    g = grad.float() * grad_scale
+    g = g.add(param, alpha=weight_decay)
+    if momentum_buf is not None:
+        if first_step:
+            momentum_buf.copy_(g)
+        else:
+            momentum_buf.mul_(momentum).add_(g, alpha=1.0 - dampening)
+        g = g.add(momentum_buf, alpha=momentum) if nesterov \
+            else momentum_buf.clone()
+    param.add_(g, alpha=-lr)
+    if param_bf16 is not None:
+        param_bf16.copy_(param.to(torch.bfloat16))
+
+
+def fused_adam(param: torch.Tensor, grad: torch.Tensor,
+               exp_avg: torch.Tensor, exp_avg_sq: torch.Tensor,
+               param_bf16: Optional[torch.Tensor] = None,
+               *, lr: float, beta1: float = 0.9, beta2: float = 0.999,
+               eps: float = 1e-8, weight_decay: float = 0.0,
+               adamw: bool = False, step: int = 1,
+               grad_scale: float = 1.0) -> None:
+    if _on_gpu(param, grad):
+        _require_ext()
+        _C.fused_adam(param, grad, exp_avg, exp_avg_sq, param_bf16, lr,
+                      beta1, beta2, eps, weight_decay, adamw, step,
+                      grad_scale)
+        return
+    g = grad.float() * grad_scale
+    if adamw:
+        param.mul_(1.0 - lr * weight_decay)
+    else:
+        g = g.add(param, alpha=weight_decay)
+    exp_avg.mul_(beta1).add_(g, alpha=1 - beta1)
+    exp_avg_sq.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+    bc1 = 1 - beta1 ** step
+    bc2 = 1 - beta2 ** step
+    denom = (exp_avg_sq / bc2).sqrt_().add_(eps)
+    param.addcdiv_(exp_avg / bc1, denom, value=-lr)
+    if param_bf16 is not None:
+        param_bf16.copy_(param.to(torch.bfloat16))
+
+
+def fused_adagrad(param: torch.Tensor, grad: torch.Tensor,
+                  state_sum: torch.Tensor, *, lr: float, eps: float = 1e-10,
+                  weight_decay: float = 0.0,
+                  grad_scale: float = 1.0) -> None:
+    if _on_gpu(param, grad):
+        _require_ext()
+        _C.fused_adagrad(param, grad, state_sum, lr, eps, weight_decay,
+                         grad_scale)
+        return
+    g = grad.float() * grad_scale
+    g = g.add(param, alpha=weight_decay)
+    state_sum.addcmul_(g, g, value=1.0)
+    param.addcdiv_(g, state_sum.sqrt().add_(eps), value=-lr)
+
+
+def fused_adadelta(param: torch.Tensor, grad: torch.Tensor,
+                   square_avg: torch.Tensor, acc_delta: torch.Tensor,
+                   *, lr: float = 1.0, rho: float = 0.9, eps: float = 1e-6,
+                   weight_decay: float = 0.0,
+                   grad_scale: float = 1.0) -> None:
+    if _on_gpu(param, grad):
+        _require_ext()
+        _C.fused_adadelta(param, grad, square_avg, acc_delta, lr, rho, eps,
+                          weight_decay, grad_scale)
+        return
+    g = grad.float() * grad_scale
+    g = g.add(param, alpha=weight_decay)
+    square_avg.mul_(rho).addcmul_(g, g, value=1 - rho)
+    dx = (acc_delta + eps).sqrt_().div_((square_avg + eps).sqrt()).mul_(g)
+    acc_delta.mul_(rho).addcmul_(dx, dx, value=1 - rho)
+    param.add_(dx, alpha=-lr)
+
+
+# ---- embedding -------------------------------------------------------------
+
+def emb_fwd(table: torch.Tensor, ids: torch.Tensor,
+            out_bf16: bool = False) -> torch.Tensor:
+    """Row gather: out[i, :] = table[ids[i], :] (ids pre-offset, flat)."""
+    if _on_gpu(table, ids):
+        _require_ext()
+        return _C.emb_fwd(table, ids, out_bf16)
+    out = table.index_select(0, ids.reshape(-1))
+    return out.to(torch.bfloat16) if out_bf16 else out
+
+
+def emb_bwd_sgd(table: torch.Tensor, ids: torch.Tensor, grad: torch.Tensor,
+                *, lr: float, scale: float = 1.0) -> None:
+    """table[ids[i], :] -= lr * scale * grad[i, :] (fused sparse update)."""
+    if _on_gpu(table, ids, grad):
+        _require_ext()
+        _C.emb_bwd_sgd(table, ids, grad.contiguous(), lr, scale)
+        return
+    table.index_add_(0, ids.reshape(-1),
+                     grad.reshape(ids.numel(), -1).float(),
+                     alpha=-lr * scale)
+
+
+def emb_bwd_dense(grad_table: torch.Tensor, ids: torch.Tensor,
+                  grad: torch.Tensor, scale: float = 1.0) -> None:
+    if _on_gpu(grad_table, ids, grad):
+        _require_ext()
+        _C.emb_bwd_dense(grad_table, ids, grad.contiguous(), scale)
+        return
+    grad_table.index_add_(0, ids.reshape(-1),
+                          grad.reshape(ids.numel(), -1).float(),
+                          alpha=scale)
+
+
+# ---- elementwise -----------------------------------------------------------
+
+def bias_relu_fwd(x: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
+    if _on_gpu(x, bias):
+        _require_ext()
+        return _C.bias_relu_fwd(x.contiguous(), bias.contiguous())
+    return torch.relu(x + bias)
+
+
+def bias_relu_bwd(dy: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+    if _on_gpu(dy, y):
+        _require_ext()
+        return _C.bias_relu_bwd(dy.contiguous(), y.contiguous())
+    return dy * (y > 0).to(dy.dtype)
+
+
+def convert_scaled(src: torch.Tensor, dst: torch.Tensor,
+                   scale: float = 1.0) -> None:
+    if _on_gpu(src, dst):
+        _require_ext()
+        _C.convert_scaled(src, dst, scale)
+        return
+    dst.copy_((src.float() * scale).to(dst.dtype))
+
+
+class BiasReLU(torch.autograd.Function):
+    """Autograd wrapper for the fused bias+ReLU epilogue."""
+
+    @staticmethod
+    def forward(ctx, x, bias):
+        y = bias_relu_fwd(x, bias)
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        dx = bias_relu_bwd(dy.contiguous(), y)
+        dims = tuple(range(dx.dim() - 1))
+        dbias = dx.sum(dim=dims)
+        return dx, dbias
+
+
+def bias_relu(x: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
+    return BiasReLU.apply(x, bias)

@@ -1,0 +1,51 @@
+// Python bindings for the tf_yarn_amd MI355X kernel library.
+#include <torch/extension.h>
+
+// fused_optimizers.hip
+void fused_sgd(torch::Tensor param, torch::Tensor grad,
+               c10::optional<torch::Tensor> momentum_buf,
+               c10::optional<torch::Tensor> param_bf16,
+               double lr, double momentum, double dampening,
+               double weight_decay, bool nesterov, bool first_step,
+               double grad_scale);
+void fused_adam(torch::Tensor param, torch::Tensor grad,
+                torch::Tensor exp_avg, torch::Tensor exp_avg_sq,
+                c10::optional<torch::Tensor> param_bf16,
+                double lr, double beta1, double beta2, double eps,
+                double weight_decay, bool adamw, int64_t step,
+                double grad_scale);
+void fused_adagrad(torch::Tensor param, torch::Tensor grad,
+                   torch::Tensor state_sum, double lr, double eps,
+                   double weight_decay, double grad_scale);
+void fused_adadelta(torch::Tensor param, torch::Tensor grad,
+                    torch::Tensor square_avg, torch::Tensor acc_delta,
+                    double lr, double rho, double eps, double weight_decay,
+                    double grad_scale);
+
+// embedding.hip
+torch::Tensor emb_fwd(torch::Tensor table, torch::Tensor ids, bool out_bf16);
+void emb_bwd_sgd(torch::Tensor table, torch::Tensor ids, torch::Tensor grad,
+                 double lr, double scale);
+void emb_bwd_dense(torch::Tensor grad_table, torch::Tensor ids,
+                   torch::Tensor grad, double scale);
+
+// elementwise.hip
+torch::Tensor bias_relu_fwd(torch::Tensor x, torch::Tensor bias);
+torch::Tensor bias_relu_bwd(torch::Tensor dy, torch::Tensor y);
+void convert_scaled(torch::Tensor src, torch::Tensor dst, double scale);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "tf_yarn_amd MI355X (gfx950) HIP kernels";
+  m.def("fused_sgd", &fused_sgd, "Fused SGD(+momentum) step");
+  m.def("fused_adam", &fused_adam, "Fused Adam/AdamW step");
+  m.def("fused_adagrad", &fused_adagrad, "Fused Adagrad step");
+  m.def("fused_adadelta", &fused_adadelta, "Fused Adadelta step");
+  m.def("emb_fwd", &emb_fwd, "Fused multi-table embedding gather");
+  m.def("emb_bwd_sgd", &emb_bwd_sgd,
+        "Fused sparse embedding grad scatter + SGD update");
+  m.def("emb_bwd_dense", &emb_bwd_dense,
+        "Sparse embedding grad scatter into dense grad table");
+  m.def("bias_relu_fwd", &bias_relu_fwd, "Fused bias+ReLU forward");
+  m.def("bias_relu_bwd", &bias_relu_bwd, "Fused ReLU backward");
+  m.def("convert_scaled", &convert_scaled, "Scaled bf16<->fp32 convert");
+}

@@ -1,0 +1,62 @@
+// Common device helpers for tf_yarn_amd CDNA4 (gfx950) kernels.
+//
+// Conventions (per the MI355X HIP guide):
+//  * wavefront = 64 lanes; block sizes are multiples of 64 (default 256)
+//  * memory-bound kernels vectorize to 16 B/lane (float4 / 8x bf16) and
+//    grid-stride with the grid capped (~256 CUs x 8 blocks)
+//  * bf16 is loaded/stored as packed ushort vectors; bf16->f32 is an exact
+//    bit shift, f32->bf16 rounds to nearest-even
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define MIYARN_BLOCK 256
+#define MIYARN_MAX_BLOCKS 2048  // 256 CUs * 8 blocks/CU
+
+static inline int miyarn_grid(int64_t n_items) {
+  int64_t blocks = (n_items + MIYARN_BLOCK - 1) / MIYARN_BLOCK;
+  if (blocks > MIYARN_MAX_BLOCKS) blocks = MIYARN_MAX_BLOCKS;
+  if (blocks < 1) blocks = 1;
+  return static_cast<int>(blocks);
+}
+
+// ---- bf16 <-> f32 ----------------------------------------------------------
+
+__device__ __forceinline__ float bf16_to_f32(unsigned short u) {
+  union { unsigned int i; float f; } v;
+  v.i = static_cast<unsigned int>(u) << 16;
+  return v.f;
+}
+
+__device__ __forceinline__ unsigned short f32_to_bf16(float f) {
+  union { float f; unsigned int i; } v;
+  v.f = f;
+  unsigned int x = v.i;
+  // round-to-nearest-even on the truncated 16 bits
+  unsigned int rounding = 0x7fffu + ((x >> 16) & 1u);
+  x += rounding;
+  return static_cast<unsigned short>(x >> 16);
+}
+
+// packed vectors: 16 B per lane
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(8))) unsigned short bf16x8;
+typedef __attribute__((ext_vector_type(4))) unsigned short bf16x4;
+
+// Generic element accessors so kernels template over float / bf16 storage.
+struct F32Io {
+  using scalar_t = float;
+  __device__ static float load(const float* p, int64_t i) { return p[i]; }
+  __device__ static void store(float* p, int64_t i, float v) { p[i] = v; }
+};
+
+struct Bf16Io {
+  using scalar_t = unsigned short;
+  __device__ static float load(const unsigned short* p, int64_t i) {
+    return bf16_to_f32(p[i]);
+  }
+  __device__ static void store(unsigned short* p, int64_t i, float v) {
+    p[i] = f32_to_bf16(v);
+  }
+};

@@ -1,0 +1,159 @@
+// Elementwise fused kernels: bias+ReLU (fwd/bwd) and bucket dtype convert.
+//
+// The MLP hot path of the wide-and-deep model: GEMMs go through
+// rocBLAS/hipBLASLt (library GEMMs), while the epilogue (bias+activation)
+// and the reducer's mixed-precision bucket casts are fused here so
+// activations make one HBM round trip instead of three
+// (guide Appendix B: element-wise — vectorize bf16 as packed shorts,
+// grid-stride, grid capped).
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include "common.h"
+
+namespace {
+
+// y = relu(x + bias), mask output for backward packed in sign of y (y==0).
+template <typename Io>
+__global__ void bias_relu_fwd_kernel(
+    const typename Io::scalar_t* __restrict__ x,
+    const typename Io::scalar_t* __restrict__ bias,
+    typename Io::scalar_t* __restrict__ y,
+    int64_t rows, int64_t cols) {
+  const int64_t total = rows * cols;
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       i < total; i += stride) {
+    const int64_t c = i % cols;
+    float v = Io::load(x, i) + Io::load(bias, c);
+    Io::store(y, i, v > 0.f ? v : 0.f);
+  }
+}
+
+// dx = dy * (y > 0)
+template <typename Io>
+__global__ void bias_relu_bwd_kernel(
+    const typename Io::scalar_t* __restrict__ dy,
+    const typename Io::scalar_t* __restrict__ y,
+    typename Io::scalar_t* __restrict__ dx,
+    int64_t total) {
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       i < total; i += stride) {
+    float g = Io::load(dy, i);
+    float yy = Io::load(y, i);
+    Io::store(dx, i, yy > 0.f ? g : 0.f);
+  }
+}
+
+__global__ void bf16_to_f32_kernel(const unsigned short* __restrict__ src,
+                                   float* __restrict__ dst, int64_t n,
+                                   float scale) {
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  const int64_t nvec = n >> 2;
+  const int64_t tid = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  for (int64_t i = tid; i < nvec; i += stride) {
+    bf16x4 v = reinterpret_cast<const bf16x4*>(src)[i];
+    f32x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o[j] = bf16_to_f32(v[j]) * scale;
+    reinterpret_cast<f32x4*>(dst)[i] = o;
+  }
+  for (int64_t i = (nvec << 2) + tid; i < n; i += stride)
+    dst[i] = bf16_to_f32(src[i]) * scale;
+}
+
+__global__ void f32_to_bf16_kernel(const float* __restrict__ src,
+                                   unsigned short* __restrict__ dst,
+                                   int64_t n, float scale) {
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  const int64_t nvec = n >> 2;
+  const int64_t tid = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  for (int64_t i = tid; i < nvec; i += stride) {
+    f32x4 v = reinterpret_cast<const f32x4*>(src)[i];
+    bf16x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o[j] = f32_to_bf16(v[j] * scale);
+    reinterpret_cast<bf16x4*>(dst)[i] = o;
+  }
+  for (int64_t i = (nvec << 2) + tid; i < n; i += stride)
+    dst[i] = f32_to_bf16(src[i] * scale);
+}
+
+}  // namespace
+
+torch::Tensor bias_relu_fwd(torch::Tensor x, torch::Tensor bias) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x must be GPU contiguous");
+  TORCH_CHECK(bias.is_cuda() && bias.is_contiguous() &&
+              bias.scalar_type() == x.scalar_type(), "bias mismatch");
+  const int64_t cols = x.size(-1);
+  const int64_t rows = x.numel() / cols;
+  TORCH_CHECK(bias.numel() == cols, "bias size mismatch");
+  auto y = torch::empty_like(x);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int grid = miyarn_grid(rows * cols);
+  if (x.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(bias_relu_fwd_kernel<F32Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream, x.data_ptr<float>(),
+                       bias.data_ptr<float>(), y.data_ptr<float>(),
+                       rows, cols);
+  } else {
+    TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "fp32/bf16 only");
+    hipLaunchKernelGGL(bias_relu_fwd_kernel<Bf16Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream,
+                       reinterpret_cast<unsigned short*>(x.data_ptr()),
+                       reinterpret_cast<unsigned short*>(bias.data_ptr()),
+                       reinterpret_cast<unsigned short*>(y.data_ptr()),
+                       rows, cols);
+  }
+  return y;
+}
+
+torch::Tensor bias_relu_bwd(torch::Tensor dy, torch::Tensor y) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && y.is_contiguous(),
+              "dy/y must be GPU contiguous");
+  TORCH_CHECK(dy.scalar_type() == y.scalar_type() &&
+              dy.numel() == y.numel(), "dy/y mismatch");
+  auto dx = torch::empty_like(dy);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int64_t total = dy.numel();
+  int grid = miyarn_grid(total);
+  if (dy.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(bias_relu_bwd_kernel<F32Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream, dy.data_ptr<float>(),
+                       y.data_ptr<float>(), dx.data_ptr<float>(), total);
+  } else {
+    TORCH_CHECK(dy.scalar_type() == torch::kBFloat16, "fp32/bf16 only");
+    hipLaunchKernelGGL(bias_relu_bwd_kernel<Bf16Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream,
+                       reinterpret_cast<unsigned short*>(dy.data_ptr()),
+                       reinterpret_cast<unsigned short*>(y.data_ptr()),
+                       reinterpret_cast<unsigned short*>(dx.data_ptr()),
+                       total);
+  }
+  return dx;
+}
+
+void convert_scaled(torch::Tensor src, torch::Tensor dst, double scale) {
+  TORCH_CHECK(src.is_cuda() && src.is_contiguous() &&
+              dst.is_cuda() && dst.is_contiguous(), "GPU contiguous only");
+  TORCH_CHECK(src.numel() == dst.numel(), "numel mismatch");
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int64_t n = src.numel();
+  int grid = miyarn_grid((n + 3) / 4);
+  if (src.scalar_type() == torch::kBFloat16 &&
+      dst.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(bf16_to_f32_kernel, dim3(grid), dim3(MIYARN_BLOCK), 0,
+                       stream,
+                       reinterpret_cast<unsigned short*>(src.data_ptr()),
+                       dst.data_ptr<float>(), n, (float)scale);
+  } else if (src.scalar_type() == torch::kFloat32 &&
+             dst.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL(f32_to_bf16_kernel, dim3(grid), dim3(MIYARN_BLOCK), 0,
+                       stream, src.data_ptr<float>(),
+                       reinterpret_cast<unsigned short*>(dst.data_ptr()),
+                       n, (float)scale);
+  } else {
+    TORCH_CHECK(false, "convert_scaled supports bf16<->fp32 only");
+  }
+}

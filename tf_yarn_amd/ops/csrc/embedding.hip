@@ -1,0 +1,251 @@
+// Fused multi-table embedding lookup + sparse-gradient apply for MI355X.
+//
+// The dominant cost of Criteo CTR models (SURVEY §7 "hard parts": sparse
+// embedding gradient reduce).  All categorical tables are concatenated into
+// one [total_rows, D] fp32 buffer; Python pre-adds per-feature row offsets
+// into the flat id tensor, so the kernels are pure row gather / row
+// scatter:
+//
+//  * emb_fwd:      out[i, :] = table[ids[i], :]          (out fp32 or bf16;
+//                  bf16 out fuses the mixed-precision cast into the gather)
+//  * emb_bwd_sgd:  table[ids[i], :] -= lr * scale * g[i, :]   (fused
+//                  backward+update, atomicAdd fp32 — no materialized
+//                  dense gradient table)
+//  * emb_bwd_dense: grad_table[ids[i], :] += scale * g[i, :]  (for
+//                  optimizers that need the dense gradient)
+//
+// Row layout keeps consecutive d contiguous, so lanes covering one row
+// coalesce; reuse across the batch is served by L2/LLC (gather guidance:
+// guide Appendix B "Scatter/gather/embedding").
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include "common.h"
+
+namespace {
+
+template <typename OIo>
+__global__ void emb_fwd_kernel(const float* __restrict__ table,
+                               const int64_t* __restrict__ ids,
+                               typename OIo::scalar_t* __restrict__ out,
+                               int64_t n_rows, int64_t dim) {
+  const int64_t dvec = dim >> 2;
+  const int64_t total = n_rows * dvec;
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       t < total; t += stride) {
+    const int64_t row = t / dvec;
+    const int64_t c4 = t - row * dvec;
+    const int64_t src = ids[row];
+    f32x4 v = reinterpret_cast<const f32x4*>(table + src * dim)[c4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      OIo::store(out, row * dim + c4 * 4 + j, v[j]);
+  }
+}
+
+template <typename GIo>
+__global__ void emb_bwd_sgd_kernel(float* __restrict__ table,
+                                   const int64_t* __restrict__ ids,
+                                   const typename GIo::scalar_t* __restrict__ g,
+                                   int64_t n_rows, int64_t dim,
+                                   float neg_lr_scale) {
+  const int64_t dvec = dim >> 2;
+  const int64_t total = n_rows * dvec;
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       t < total; t += stride) {
+    const int64_t row = t / dvec;
+    const int64_t c4 = t - row * dvec;
+    float* dst = table + ids[row] * dim + c4 * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float gv = GIo::load(g, row * dim + c4 * 4 + j);
+      atomicAdd(dst + j, neg_lr_scale * gv);
+    }
+  }
+}
+
+template <typename GIo>
+__global__ void emb_bwd_dense_kernel(
+    float* __restrict__ grad_table,
+    const int64_t* __restrict__ ids,
+    const typename GIo::scalar_t* __restrict__ g,
+    int64_t n_rows, int64_t dim, float scale) {
+  const int64_t dvec = dim >> 2;
+  const int64_t total = n_rows * dvec;
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       t < total; t += stride) {
+    const int64_t row = t / dvec;
+    const int64_t c4 = t - row * dvec;
+    float* dst = grad_table + ids[row] * dim + c4 * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float gv = GIo::load(g, row * dim + c4 * 4 + j);
+      atomicAdd(dst + j, scale * gv);
+    }
+  }
+}
+
+// Scalar variants for dim % 4 != 0 (e.g. the wide part's dim-1 tables).
+template <typename OIo>
+__global__ void emb_fwd_scalar_kernel(const float* __restrict__ table,
+                                      const int64_t* __restrict__ ids,
+                                      typename OIo::scalar_t* __restrict__ out,
+                                      int64_t n_rows, int64_t dim) {
+  const int64_t total = n_rows * dim;
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       t < total; t += stride) {
+    const int64_t row = t / dim;
+    const int64_t c = t - row * dim;
+    OIo::store(out, t, table[ids[row] * dim + c]);
+  }
+}
+
+template <typename GIo>
+__global__ void emb_scatter_scalar_kernel(
+    float* __restrict__ table, const int64_t* __restrict__ ids,
+    const typename GIo::scalar_t* __restrict__ g,
+    int64_t n_rows, int64_t dim, float alpha) {
+  const int64_t total = n_rows * dim;
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       t < total; t += stride) {
+    const int64_t row = t / dim;
+    const int64_t c = t - row * dim;
+    atomicAdd(table + ids[row] * dim + c, alpha * GIo::load(g, t));
+  }
+}
+
+void check_emb(const torch::Tensor& table, const torch::Tensor& ids,
+               int64_t dim) {
+  TORCH_CHECK(table.is_cuda() && table.is_contiguous() &&
+              table.scalar_type() == torch::kFloat32,
+              "table must be a contiguous fp32 GPU tensor");
+  TORCH_CHECK(ids.is_cuda() && ids.is_contiguous() &&
+              ids.scalar_type() == torch::kInt64,
+              "ids must be contiguous int64 on GPU");
+}
+
+}  // namespace
+
+torch::Tensor emb_fwd(torch::Tensor table, torch::Tensor ids,
+                      bool out_bf16) {
+  const int64_t dim = table.size(1);
+  const int64_t n = ids.numel();
+  check_emb(table, ids, dim);
+  auto out = torch::empty(
+      {n, dim}, table.options().dtype(
+          out_bf16 ? torch::kBFloat16 : torch::kFloat32));
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const bool vec = (dim % 4 == 0);
+  int grid = miyarn_grid(n * (vec ? dim / 4 : dim));
+  if (out_bf16) {
+    auto* optr = reinterpret_cast<unsigned short*>(out.data_ptr());
+    if (vec)
+      hipLaunchKernelGGL(emb_fwd_kernel<Bf16Io>, dim3(grid),
+                         dim3(MIYARN_BLOCK), 0, stream,
+                         table.data_ptr<float>(), ids.data_ptr<int64_t>(),
+                         optr, n, dim);
+    else
+      hipLaunchKernelGGL(emb_fwd_scalar_kernel<Bf16Io>, dim3(grid),
+                         dim3(MIYARN_BLOCK), 0, stream,
+                         table.data_ptr<float>(), ids.data_ptr<int64_t>(),
+                         optr, n, dim);
+  } else {
+    if (vec)
+      hipLaunchKernelGGL(emb_fwd_kernel<F32Io>, dim3(grid),
+                         dim3(MIYARN_BLOCK), 0, stream,
+                         table.data_ptr<float>(), ids.data_ptr<int64_t>(),
+                         out.data_ptr<float>(), n, dim);
+    else
+      hipLaunchKernelGGL(emb_fwd_scalar_kernel<F32Io>, dim3(grid),
+                         dim3(MIYARN_BLOCK), 0, stream,
+                         table.data_ptr<float>(), ids.data_ptr<int64_t>(),
+                         out.data_ptr<float>(), n, dim);
+  }
+  return out;
+}
+
+void emb_bwd_sgd(torch::Tensor table, torch::Tensor ids, torch::Tensor grad,
+                 double lr, double scale) {
+  const int64_t dim = table.size(1);
+  const int64_t n = ids.numel();
+  check_emb(table, ids, dim);
+  TORCH_CHECK(grad.is_cuda() && grad.is_contiguous() &&
+              grad.numel() == n * dim, "grad shape mismatch");
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const bool vec = (dim % 4 == 0);
+  int grid = miyarn_grid(n * (vec ? dim / 4 : dim));
+  float nls = (float)(-lr * scale);
+  if (grad.scalar_type() == torch::kFloat32) {
+    if (vec)
+      hipLaunchKernelGGL(emb_bwd_sgd_kernel<F32Io>, dim3(grid),
+                         dim3(MIYARN_BLOCK), 0, stream,
+                         table.data_ptr<float>(), ids.data_ptr<int64_t>(),
+                         grad.data_ptr<float>(), n, dim, nls);
+    else
+      hipLaunchKernelGGL(emb_scatter_scalar_kernel<F32Io>, dim3(grid),
+                         dim3(MIYARN_BLOCK), 0, stream,
+                         table.data_ptr<float>(), ids.data_ptr<int64_t>(),
+                         grad.data_ptr<float>(), n, dim, nls);
+  } else {
+    TORCH_CHECK(grad.scalar_type() == torch::kBFloat16,
+                "grad must be fp32 or bf16");
+    auto* gptr = reinterpret_cast<unsigned short*>(grad.data_ptr());
+    if (vec)
+      hipLaunchKernelGGL(emb_bwd_sgd_kernel<Bf16Io>, dim3(grid),
+                         dim3(MIYARN_BLOCK), 0, stream,
+                         table.data_ptr<float>(), ids.data_ptr<int64_t>(),
+                         gptr, n, dim, nls);
+    else
+      hipLaunchKernelGGL(emb_scatter_scalar_kernel<Bf16Io>, dim3(grid),
+                         dim3(MIYARN_BLOCK), 0, stream,
+                         table.data_ptr<float>(), ids.data_ptr<int64_t>(),
+                         gptr, n, dim, nls);
+  }
+}
+
+void emb_bwd_dense(torch::Tensor grad_table, torch::Tensor ids,
+                   torch::Tensor grad, double scale) {
+  const int64_t dim = grad_table.size(1);
+  const int64_t n = ids.numel();
+  check_emb(grad_table, ids, dim);
+  TORCH_CHECK(grad.is_cuda() && grad.is_contiguous() &&
+              grad.numel() == n * dim, "grad shape mismatch");
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const bool vec = (dim % 4 == 0);
+  int grid = miyarn_grid(n * (vec ? dim / 4 : dim));
+  if (grad.scalar_type() == torch::kFloat32) {
+    if (vec)
+      hipLaunchKernelGGL(emb_bwd_dense_kernel<F32Io>, dim3(grid),
+                         dim3(MIYARN_BLOCK), 0, stream,
+                         grad_table.data_ptr<float>(),
+                         ids.data_ptr<int64_t>(), grad.data_ptr<float>(),
+                         n, dim, (float)scale);
+    else
+      hipLaunchKernelGGL(emb_scatter_scalar_kernel<F32Io>, dim3(grid),
+                         dim3(MIYARN_BLOCK), 0, stream,
+                         grad_table.data_ptr<float>(),
+                         ids.data_ptr<int64_t>(), grad.data_ptr<float>(),
+                         n, dim, (float)scale);
+  } else {
+    TORCH_CHECK(grad.scalar_type() == torch::kBFloat16,
+                "grad must be fp32 or bf16");
+    auto* gptr = reinterpret_cast<unsigned short*>(grad.data_ptr());
+    if (vec)
+      hipLaunchKernelGGL(emb_bwd_dense_kernel<Bf16Io>, dim3(grid),
+                         dim3(MIYARN_BLOCK), 0, stream,
+                         grad_table.data_ptr<float>(),
+                         ids.data_ptr<int64_t>(), gptr, n, dim,
+                         (float)scale);
+    else
+      hipLaunchKernelGGL(emb_scatter_scalar_kernel<Bf16Io>, dim3(grid),
+                         dim3(MIYARN_BLOCK), 0, stream,
+                         grad_table.data_ptr<float>(),
+                         ids.data_ptr<int64_t>(), gptr, n, dim,
+                         (float)scale);
+  }
+}

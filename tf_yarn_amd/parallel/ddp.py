@@ -105,7 +105,11 @@ class BucketedDataParallel(nn.Module):
     # -- construction -------------------------------------------------------
 
     def _build_buckets(self) -> None:
-        params = [p for p in self.module.parameters() if p.requires_grad]
+        # Params flagged _miyarn_sparse (embedding tables) sync through the
+        # sparse allgather path, not dense bucket allreduce.
+        params = [p for p in self.module.parameters()
+                  if p.requires_grad
+                  and not getattr(p, "_miyarn_sparse", False)]
         # Reverse registration order approximates backward completion order.
         params = list(reversed(params))
         groups: List[List[nn.Parameter]] = []
@@ -159,6 +163,10 @@ class BucketedDataParallel(nn.Module):
                 bucket.buffer.zero_()
             for buf in self.module.buffers():
                 dist.broadcast(buf, src=0, group=self.process_group)
+            for p in self.module.parameters():
+                if getattr(p, "_miyarn_sparse", False):
+                    dist.broadcast(p.detach(), src=0,
+                                   group=self.process_group)
         # Grads for non-bucketed params (requires_grad=False) never sync.
 
     # -- backward machinery --------------------------------------------------
