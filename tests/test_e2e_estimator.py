@@ -1,0 +1,117 @@
+"""End-to-end estimator-flavor runs on CPU: PS strategy, ring allreduce,
+and the full topology with evaluator + tensorboard side tasks."""
+
+import os
+import sys
+
+import cloudpickle
+import pytest
+import torch
+from torch import nn
+
+from tf_yarn_amd import TaskSpec, run_on_yarn
+from tf_yarn_amd.estimator import run_on_yarn as est_run_on_yarn
+
+cloudpickle.register_pickle_by_value(sys.modules[__name__])
+
+
+def _experiment_fn(model_dir, max_steps=12):
+    import torch
+    from torch import nn
+
+    from tf_yarn_amd.estimator import (Estimator, EvalSpec, RunConfig,
+                                       TrainSpec)
+    from tf_yarn_amd.estimator.experiment import Experiment
+
+    def module_fn():
+        torch.manual_seed(0)
+        return nn.Sequential(nn.Linear(4, 8), nn.ReLU(), nn.Linear(8, 2))
+
+    def optimizer_fn(params):
+        return torch.optim.SGD(params, lr=0.05)
+
+    def loss_fn(out, labels):
+        return nn.functional.cross_entropy(out, labels.long())
+
+    def input_fn():
+        torch.manual_seed(3)
+        for _ in range(50):
+            x = torch.randn(16, 4)
+            yield x, (x.sum(dim=1) > 0).long()
+
+    est = Estimator(module_fn, optimizer_fn, loss_fn,
+                    model_dir=model_dir,
+                    config=RunConfig(save_checkpoints_steps=6),
+                    device="cpu")
+    return Experiment(
+        est,
+        TrainSpec(input_fn, max_steps=max_steps),
+        EvalSpec(input_fn, steps=3, throttle_secs=0))
+
+
+@pytest.mark.timeout(240)
+def test_ps_strategy_cpu(tmp_path):
+    """1 chief + 1 ps + 2 workers over gloo: async push/pull training
+    (BASELINE config 2 topology, CPU plumbing)."""
+    model_dir = str(tmp_path / "model")
+    from functools import partial
+    metrics = est_run_on_yarn(
+        partial(_experiment_fn, model_dir),
+        {
+            "chief": TaskSpec(memory=512, vcores=1),
+            "ps": TaskSpec(memory=512, vcores=1, instances=1),
+            "worker": TaskSpec(memory=512, vcores=1, instances=2),
+        },
+        base_dir=str(tmp_path / "app"),
+    )
+    assert metrics is not None
+    # chief wrote the model.ckpt layout
+    assert os.path.exists(os.path.join(model_dir, "checkpoint"))
+    names = os.listdir(model_dir)
+    assert any(n.startswith("model.ckpt-") for n in names)
+
+
+@pytest.mark.timeout(240)
+def test_allreduce_cpu(tmp_path):
+    """chief + 1 worker ring-allreduce via the allred task module
+    (the Horovod-gloo path re-implemented on the framework engine)."""
+    model_dir = str(tmp_path / "model")
+    from functools import partial
+    metrics = est_run_on_yarn(
+        partial(_experiment_fn, model_dir),
+        {
+            "chief": TaskSpec(memory=512, vcores=1),
+            "worker": TaskSpec(memory=512, vcores=1, instances=1),
+        },
+        custom_task_module="tf_yarn_amd.estimator.tasks.allred_task",
+        base_dir=str(tmp_path / "app"),
+    )
+    assert metrics is not None
+    assert metrics.total_training_duration is not None
+    assert os.path.exists(os.path.join(model_dir, "checkpoint"))
+
+
+@pytest.mark.timeout(300)
+def test_full_topology_with_side_tasks(tmp_path):
+    """chief + evaluator + tensorboard co-tasks (BASELINE config 5 shape):
+    evaluator evaluates the chief's checkpoints; tensorboard advertises a
+    URL and exits after the stop barrier."""
+    model_dir = str(tmp_path / "model")
+    from functools import partial
+    metrics = est_run_on_yarn(
+        partial(_experiment_fn, model_dir),
+        {
+            "chief": TaskSpec(memory=512, vcores=1),
+            "evaluator": TaskSpec(memory=512, vcores=1),
+            "tensorboard": TaskSpec(memory=512, vcores=1,
+                                    tb_termination_timeout_seconds=1,
+                                    tb_model_dir=model_dir),
+        },
+        base_dir=str(tmp_path / "app"),
+    )
+    assert metrics is not None
+    # evaluator produced eval events for the final checkpoint
+    eval_dir = os.path.join(model_dir, "eval")
+    assert os.path.isdir(eval_dir)
+    from tf_yarn_amd.estimator.estimator import evaluated_steps
+    assert 12 in evaluated_steps(eval_dir)

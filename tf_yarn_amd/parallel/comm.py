@@ -35,17 +35,23 @@ def init_process_group(rank: int,
                        device: Optional[torch.device | str] = None,
                        kv_client: Optional[KVClient] = None,
                        timeout_minutes: float = 30,
-                       group_name: str = "default") -> None:
+                       group_name: str = "default",
+                       need_subgroups: bool = False) -> None:
     """Initialize torch.distributed through the KV control plane.
 
     Falls back to env:// (MASTER_ADDR/MASTER_PORT, the reference's
-    ``worker.py:101`` contract) when no KV client is available, e.g. under
-    ``torch.distributed.run``.
+    ``worker.py:101`` contract) when no KV client is available (e.g. under
+    ``torch.distributed.run``) or when the caller needs ``dist.new_group``
+    sub-groups (``need_subgroups=True``): c10d cannot derive sub-group
+    stores from a Python Store subclass, so the PS pair-group path
+    rendezvouses on a TCPStore at the KV-elected master address instead.
     """
     if backend is None:
         backend = get_backend_for_device(device or (
             "cuda" if torch.cuda.is_available() else "cpu"))
     timeout = timedelta(minutes=timeout_minutes)
+    if need_subgroups:
+        kv_client = None
     if kv_client is not None:
         store = KVRendezvousStore(kv_client, prefix=f"c10d/{group_name}/",
                                   timeout=timeout)

@@ -1,0 +1,270 @@
+"""Parameter-server data plane on RCCL point-to-point over xGMI.
+
+MI355X-native replacement for the reference's TF gRPC
+ParameterServerStrategy: parameter shards live on the ps
+tasks' GPUs; workers push gradients and pull fresh weights with
+``torch.distributed`` send/recv (RCCL p2p over xGMI for GPU ranks, gloo for
+CPU plumbing), and the ps applies a fused HIP optimizer step per push —
+asynchronous semantics: no barrier between workers, each push/pull pair is
+independent (PS staleness model preserved).
+
+Wire protocol per (worker, ps) pair — each pair has its own process group
+(own RCCL communicator), so the ps can serve workers from one thread per
+worker without cross-thread communicator sharing:
+
+    worker -> ps : header [1] int64  (1 = push/pull, 0 = goodbye)
+    worker -> ps : grad shard flat   (only if header == 1)
+    ps     -> worker : weight shard flat
+
+Sharding: parameters are partitioned greedily by size across ps instances;
+both sides derive the identical partition from the module structure.
+Initial weights flow chief -> ps at setup so every shard starts from the
+chief's random init.
+"""
+
+from __future__ import annotations
+
+import logging
+import threading
+from typing import Callable, Dict, List, Optional, Sequence, Tuple
+
+import torch
+import torch.distributed as dist
+
+logger = logging.getLogger(__name__)
+
+
+def shard_parameters(shapes: List[Tuple[int, ...]],
+                     n_shards: int) -> List[List[int]]:
+    """Greedy partition of param indices by numel across shards.
+    Deterministic: both workers and ps compute the same layout."""
+    sizes = [(int(torch.tensor(s).prod()) if len(s) else 1, i)
+             for i, s in enumerate(shapes)]
+    shards: List[List[int]] = [[] for _ in range(n_shards)]
+    loads = [0] * n_shards
+    for numel, i in sorted(sizes, reverse=True):
+        j = loads.index(min(loads))
+        shards[j].append(i)
+        loads[j] += numel
+    return [sorted(s) for s in shards]
+
+
+class _ShardLayout:
+    def __init__(self, params: Sequence[torch.Tensor], n_shards: int):
+        self.shapes = [tuple(p.shape) for p in params]
+        self.numels = [p.numel() for p in params]
+        self.assignment = shard_parameters(self.shapes, n_shards)
+        self.shard_numel = [
+            sum(self.numels[i] for i in idxs) for idxs in self.assignment]
+
+    def pack(self, shard: int, tensors: Sequence[torch.Tensor],
+             out: torch.Tensor) -> torch.Tensor:
+        offset = 0
+        for i in self.assignment[shard]:
+            n = self.numels[i]
+            out[offset:offset + n].copy_(tensors[i].reshape(-1))
+            offset += n
+        return out
+
+    def unpack(self, shard: int, flat: torch.Tensor,
+               tensors: Sequence[torch.Tensor]) -> None:
+        offset = 0
+        for i in self.assignment[shard]:
+            n = self.numels[i]
+            tensors[i].reshape(-1).copy_(flat[offset:offset + n])
+            offset += n
+
+
+class PsTopology:
+    """Global rank layout: training procs first (cluster-task order), then
+    ps procs.  Built identically on every process from cluster_tasks."""
+
+    def __init__(self, cluster_tasks, my_task_type: str, my_task_id: int,
+                 local_rank: int = 0):
+        training = [t for t in cluster_tasks if t.type in ("chief",
+                                                           "worker")]
+        ps_tasks = [t for t in cluster_tasks if t.type == "ps"]
+        self.n_workers = sum(t.nb_proc for t in training)
+        self.n_ps = len(ps_tasks)
+        self.world_size = self.n_workers + self.n_ps
+        rank = 0
+        self.rank = None
+        for t in training:
+            if t.type == my_task_type and t.id == my_task_id:
+                self.rank = rank + local_rank
+            rank += t.nb_proc
+        for k, t in enumerate(ps_tasks):
+            if t.type == "ps" and my_task_type == "ps" \
+                    and t.id == my_task_id:
+                self.rank = rank + k
+        if self.rank is None:
+            raise ValueError(
+                f"{my_task_type}:{my_task_id} not in PS topology")
+        self.ps_ranks = list(range(self.n_workers, self.world_size))
+        self.worker_ranks = list(range(self.n_workers))
+        self.is_ps = self.rank >= self.n_workers
+
+
+def build_pair_groups(topo: PsTopology):
+    """Create one process group per (worker, ps) pair; every rank calls
+    new_group for every pair, in identical order (c10d requirement)."""
+    groups = {}
+    for w in topo.worker_ranks:
+        for s in topo.ps_ranks:
+            groups[(w, s)] = dist.new_group([w, s])
+    return groups
+
+
+HEADER_PUSH = 1
+HEADER_BYE = 0
+
+
+class PsShardServer:
+    """Runs on a ps task: holds its shard + optimizer state; serves one
+    thread per worker.  ``optimizer_step(shard_flat, grad_flat, lock)``
+    applies the fused update in place."""
+
+    def __init__(self, topo: PsTopology, layout: _ShardLayout,
+                 pair_groups: Dict, device: str,
+                 optimizer_step: Callable[[torch.Tensor, torch.Tensor],
+                                          None]):
+        self.topo = topo
+        self.layout = layout
+        self.pair_groups = pair_groups
+        self.device = device
+        self.shard_index = topo.rank - topo.n_workers
+        self.numel = layout.shard_numel[self.shard_index]
+        self.shard = torch.zeros(self.numel, device=device)
+        self.optimizer_step = optimizer_step
+        self._lock = threading.Lock()
+
+    def receive_initial(self, src_rank: int = 0) -> None:
+        """Chief pushes initial shard values (worker side:
+        :meth:`PsWorkerChannel.send_initial`)."""
+        group = self.pair_groups[(src_rank, self.topo.rank)]
+        dist.recv(self.shard, src=src_rank, group=group)
+        logger.info("ps shard %d: received %d initial weights",
+                    self.shard_index, self.numel)
+
+    def serve(self) -> None:
+        """Blockingly serve all workers until each says goodbye."""
+        threads = [threading.Thread(target=self._serve_worker, args=(w,),
+                                    name=f"ps-serve-w{w}", daemon=True)
+                   for w in self.topo.worker_ranks]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        logger.info("ps shard %d: all workers done", self.shard_index)
+
+    def _serve_worker(self, w: int) -> None:
+        group = self.pair_groups[(w, self.topo.rank)]
+        header = torch.zeros(1, dtype=torch.int64, device=self.device)
+        grad = torch.zeros(self.numel, device=self.device)
+        while True:
+            dist.recv(header, src=w, group=group)
+            if int(header.item()) == HEADER_BYE:
+                return
+            dist.recv(grad, src=w, group=group)
+            with self._lock:
+                self.optimizer_step(self.shard, grad)
+                # snapshot under the lock so the reply is consistent
+                reply = self.shard.clone()
+            dist.send(reply, dst=w, group=group)
+
+
+class PsWorkerChannel:
+    """Runs on a training proc: push grads / pull weights per shard."""
+
+    def __init__(self, topo: PsTopology, layout: _ShardLayout,
+                 pair_groups: Dict, device: str,
+                 params: Sequence[torch.Tensor]):
+        self.topo = topo
+        self.layout = layout
+        self.pair_groups = pair_groups
+        self.device = device
+        self.params = list(params)
+        self._grad_bufs = [
+            torch.zeros(n, device=device) for n in layout.shard_numel]
+        self._weight_bufs = [
+            torch.zeros(n, device=device) for n in layout.shard_numel]
+        self._header = torch.ones(1, dtype=torch.int64, device=device)
+
+    def send_initial(self) -> None:
+        """Chief only: push initial weights to every shard."""
+        with torch.no_grad():
+            for k, s in enumerate(self.topo.ps_ranks):
+                self.layout.pack(k, [p.detach() for p in self.params],
+                                 self._weight_bufs[k])
+                dist.send(self._weight_bufs[k], dst=s,
+                          group=self.pair_groups[(self.topo.rank, s)])
+
+    @torch.no_grad()
+    def push_pull(self) -> None:
+        """Push p.grad, apply on ps, pull fresh weights (async PS step)."""
+        grads = [p.grad if p.grad is not None
+                 else torch.zeros_like(p) for p in self.params]
+        for k, s in enumerate(self.topo.ps_ranks):
+            group = self.pair_groups[(self.topo.rank, s)]
+            self.layout.pack(k, grads, self._grad_bufs[k])
+            dist.send(self._header, dst=s, group=group)
+            dist.send(self._grad_bufs[k], dst=s, group=group)
+        for k, s in enumerate(self.topo.ps_ranks):
+            group = self.pair_groups[(self.topo.rank, s)]
+            dist.recv(self._weight_bufs[k], src=s, group=group)
+            self.layout.unpack(k, self._weight_bufs[k], self.params)
+
+    def goodbye(self) -> None:
+        bye = torch.zeros(1, dtype=torch.int64, device=self.device)
+        for s in self.topo.ps_ranks:
+            dist.send(bye, dst=s, group=self.pair_groups[(self.topo.rank,
+                                                          s)])
+
+
+def make_sgd_step(lr: float,
+                  momentum: float = 0.0) -> Callable:
+    """Fused server-side SGD apply (HIP kernel on GPU shards)."""
+    from tf_yarn_amd import ops
+    state = {}
+
+    def step(shard: torch.Tensor, grad: torch.Tensor) -> None:
+        mom = None
+        if momentum != 0.0:
+            if "m" not in state:
+                state["m"] = torch.zeros_like(shard)
+            mom = state["m"]
+        first = not state.get("stepped", False)
+        state["stepped"] = True
+        ops.fused_sgd(shard, grad, mom, None, lr=lr, momentum=momentum,
+                      first_step=first)
+
+    return step
+
+
+def make_adagrad_step(lr: float, eps: float = 1e-10) -> Callable:
+    from tf_yarn_amd import ops
+    state = {}
+
+    def step(shard: torch.Tensor, grad: torch.Tensor) -> None:
+        if "sum" not in state:
+            state["sum"] = torch.zeros_like(shard)
+        ops.fused_adagrad(shard, grad, state["sum"], lr=lr, eps=eps)
+
+    return step
+
+
+def make_adam_step(lr: float, beta1: float = 0.9, beta2: float = 0.999,
+                   eps: float = 1e-8) -> Callable:
+    from tf_yarn_amd import ops
+    state = {"step": 0}
+
+    def step(shard: torch.Tensor, grad: torch.Tensor) -> None:
+        if "m" not in state:
+            state["m"] = torch.zeros_like(shard)
+            state["v"] = torch.zeros_like(shard)
+        state["step"] += 1
+        ops.fused_adam(shard, grad, state["m"], state["v"], None, lr=lr,
+                       beta1=beta1, beta2=beta2, eps=eps,
+                       step=state["step"])
+
+    return step
