@@ -31,7 +31,7 @@ from tf_yarn_amd.models.wide_deep import WideAndDeep  # noqa: E402
 from tf_yarn_amd.ops.optim import FusedSGD  # noqa: E402
 from tf_yarn_amd.parallel.ddp import BucketedDataParallel  # noqa: E402
 
-PER_GPU_BATCH = 16384
+PER_GPU_BATCH = 65536
 TABLE_ROWS_PER_FEATURE = 1_000_000
 EMBEDDING_DIM = 16
 HIDDEN = (1024, 512, 256)

@@ -190,3 +190,71 @@ def test_wide_deep_step_gpu_bf16():
         losses.append(loss.item())
     assert all(torch.isfinite(torch.tensor(losses)))
     assert losses[-1] < losses[0], f"loss did not decrease: {losses}"
+
+
+@requires_gpu
+def test_fused_sgd_mt_matches_per_tensor():
+    torch.manual_seed(8)
+    sizes = [437, 1024 * 100, 3, 64 * 64]
+    ps1 = [torch.randn(n).cuda() for n in sizes]
+    ps2 = [p.clone() for p in ps1]
+    gs = [torch.randn(n).cuda().to(torch.bfloat16) for n in sizes]
+    ms1 = [torch.randn(n).cuda().abs() for n in sizes]
+    ms2 = [m.clone() for m in ms1]
+    kwargs = dict(lr=0.1, momentum=0.9, weight_decay=0.01,
+                  first_step=False)
+    for p, g, m in zip(ps1, gs, ms1):
+        ops.fused_sgd(p, g, m, None, **kwargs)
+    ops.fused_sgd_mt(ps2, gs, ms2, None, **kwargs)
+    for p1, p2 in zip(ps1, ps2):
+        assert torch.equal(p1, p2)
+    for m1, m2 in zip(ms1, ms2):
+        assert torch.equal(m1, m2)
+
+
+@requires_gpu
+def test_fused_sgd_mt_many_tensors_chunking():
+    torch.manual_seed(9)
+    n_t = 30  # crosses the MT_MAX=24 per-launch boundary
+    ps = [torch.randn(100 + i).cuda() for i in range(n_t)]
+    refs = [p.clone() for p in ps]
+    gs = [torch.randn_like(p) for p in ps]
+    ops.fused_sgd_mt(ps, gs, None, None, lr=0.5)
+    for p, r, g in zip(ps, refs, gs):
+        assert torch.allclose(p, r - 0.5 * g, atol=1e-6)
+
+
+@requires_gpu
+def test_bias_relu_bwd_db_matches_reference():
+    import tf_yarn_amd.ops._C as C
+    torch.manual_seed(10)
+    for dtype in (torch.float32, torch.bfloat16):
+        y = torch.relu(torch.randn(1024, 513)).cuda().to(dtype)
+        dy = torch.randn(1024, 513).cuda().to(dtype)
+        dx, dbias = C.bias_relu_bwd_db(dy.contiguous(), y.contiguous())
+        ref_dx = dy.float() * (y.float() > 0)
+        assert torch.allclose(dx.float(), ref_dx.to(dtype).float(),
+                              atol=1e-2)
+        ref_db = ref_dx.to(dtype).float().sum(dim=0)
+        assert torch.allclose(dbias, ref_db, atol=0.5, rtol=1e-2)
+
+
+@requires_gpu
+def test_emb_gather_scatter_sum_gpu():
+    torch.manual_seed(11)
+    table = torch.randn(500, 1).cuda()
+    ids = torch.randint(0, 500, (128, 26)).cuda()
+    out = ops.emb_gather_sum(table, ids)
+    ref = table.reshape(-1).index_select(
+        0, ids.reshape(-1)).reshape(128, 26).sum(dim=1)
+    assert torch.allclose(out, ref, atol=1e-4)
+    out16 = ops.emb_gather_sum(table, ids, out_bf16=True)
+    assert out16.dtype == torch.bfloat16
+
+    grad = torch.randn(128).cuda()
+    table2 = table.clone()
+    ops.emb_scatter_sum(table2, ids, grad, alpha=-0.1)
+    ref2 = table.clone()
+    expanded = grad.reshape(-1, 1).expand(-1, 26).reshape(-1)
+    ref2.reshape(-1).index_add_(0, ids.reshape(-1), expanded, alpha=-0.1)
+    assert torch.allclose(table2, ref2, atol=1e-3)

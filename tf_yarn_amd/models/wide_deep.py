@@ -125,6 +125,80 @@ class SparseEmbedding(nn.Module):
         self._sink.clear()
 
 
+class _GatherSum(torch.autograd.Function):
+    """Wide-part fused gather-sum: out[b] = sum_f table[ids[b,f]]."""
+
+    @staticmethod
+    def forward(ctx, table, flat_ids, batch, out_bf16, sink):
+        out = ops.emb_gather_sum(
+            table, flat_ids.reshape(batch, -1), out_bf16)
+        ctx.sink = sink
+        ctx.save_for_backward(flat_ids)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        (flat_ids,) = ctx.saved_tensors
+        ctx.sink.append((flat_ids, grad_out.contiguous()))
+        return None, None, None, None, None
+
+
+class WideScalarEmbedding(nn.Module):
+    """Per-id scalar weights with fused gather-sum forward and
+    scatter-add sparse update (the wide part of wide-and-deep)."""
+
+    def __init__(self, table_sizes: List[int], out_bf16: bool = False,
+                 init_std: float = 0.01):
+        super().__init__()
+        self.table_sizes = list(table_sizes)
+        self.out_bf16 = out_bf16
+        total = sum(table_sizes)
+        offsets = torch.tensor(
+            [0] + list(torch.cumsum(torch.tensor(table_sizes), 0)[:-1]),
+            dtype=torch.int64)
+        self.register_buffer("offsets", offsets, persistent=True)
+        self.weight = nn.Parameter(torch.randn(total, 1) * init_std)
+        self.weight._miyarn_sparse = True
+        self._sink: List[Tuple[torch.Tensor, torch.Tensor]] = []
+
+    def forward(self, ids: torch.Tensor) -> torch.Tensor:
+        b = ids.shape[0]
+        flat = (ids + self.offsets.unsqueeze(0)).reshape(-1)
+        return _GatherSum.apply(self.weight, flat, b, self.out_bf16,
+                                self._sink)
+
+    def pending_grads(self):
+        return self._sink
+
+    @torch.no_grad()
+    def apply_sparse_updates(self, lr: float, process_group=None) -> None:
+        if not self._sink:
+            return
+        world = (dist.get_world_size(process_group)
+                 if dist.is_available() and dist.is_initialized() else 1)
+        for flat_ids, grad in self._sink:
+            if world > 1:
+                n = flat_ids.numel()
+                b = grad.numel()
+                all_ids = torch.empty(n * world, dtype=flat_ids.dtype,
+                                      device=flat_ids.device)
+                all_grads = torch.empty(b * world, dtype=grad.dtype,
+                                        device=grad.device)
+                dist.all_gather_into_tensor(all_ids, flat_ids,
+                                            group=process_group)
+                dist.all_gather_into_tensor(all_grads, grad.reshape(-1),
+                                            group=process_group)
+            else:
+                all_ids, all_grads = flat_ids, grad.reshape(-1)
+            ops.emb_scatter_sum(self.weight.data,
+                                all_ids.reshape(all_grads.numel(), -1),
+                                all_grads, alpha=-lr / world)
+        self._sink.clear()
+
+    def clear_pending(self) -> None:
+        self._sink.clear()
+
+
 class FusedLinearReLU(nn.Module):
     """Linear (library GEMM) + fused bias+ReLU epilogue kernel."""
 
@@ -157,8 +231,8 @@ class WideAndDeep(nn.Module):
         self.deep_embedding = SparseEmbedding(table_sizes, embedding_dim,
                                               out_bf16=bf16)
         # Wide part: one scalar weight per categorical id (linear-in-one-hot)
-        self.wide_embedding = SparseEmbedding(table_sizes, 1, out_bf16=bf16,
-                                              init_std=0.01)
+        self.wide_embedding = WideScalarEmbedding(table_sizes,
+                                                  out_bf16=bf16)
         self.wide_dense = nn.Linear(dense_dim, 1)
         layers: List[nn.Module] = []
         in_dim = dense_dim + len(table_sizes) * embedding_dim
@@ -179,7 +253,7 @@ class WideAndDeep(nn.Module):
         deep_in = torch.cat(
             [dense, self.deep_embedding(sparse_ids)], dim=1)
         deep_out = self.head(self.mlp(deep_in))
-        wide_out = (self.wide_embedding(sparse_ids).sum(dim=1, keepdim=True)
+        wide_out = (self.wide_embedding(sparse_ids).unsqueeze(1)
                     + self.wide_dense(dense))
         return (deep_out + wide_out).squeeze(1)
 

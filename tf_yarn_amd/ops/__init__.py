@@ -65,6 +65,29 @@ def fused_sgd(param: torch.Tensor, grad: torch.Tensor,
         param_bf16.copy_(param.to(torch.bfloat16))
 
 
+def fused_sgd_mt(params, grads, momentum_bufs=None, params_bf16=None,
+                 *, lr: float, momentum: float = 0.0,
+                 dampening: float = 0.0, weight_decay: float = 0.0,
+                 nesterov: bool = False, first_step: bool = False,
+                 grad_scale: float = 1.0) -> None:
+    """Multi-tensor SGD: one kernel launch per <=24 tensors."""
+    if params and params[0].is_cuda:
+        _require_ext()
+        _C.fused_sgd_mt(list(params), list(grads),
+                        list(momentum_bufs) if momentum_bufs else [],
+                        list(params_bf16) if params_bf16 else [],
+                        lr, momentum, dampening, weight_decay, nesterov,
+                        first_step, grad_scale)
+        return
+    for i, (p, g) in enumerate(zip(params, grads)):
+        fused_sgd(p, g,
+                  momentum_bufs[i] if momentum_bufs else None,
+                  params_bf16[i] if params_bf16 else None,
+                  lr=lr, momentum=momentum, dampening=dampening,
+                  weight_decay=weight_decay, nesterov=nesterov,
+                  first_step=first_step, grad_scale=grad_scale)
+
+
 def fused_adam(param: torch.Tensor, grad: torch.Tensor,
                exp_avg: torch.Tensor, exp_avg_sq: torch.Tensor,
                param_bf16: Optional[torch.Tensor] = None,
@@ -186,8 +209,36 @@ def convert_scaled(src: torch.Tensor, dst: torch.Tensor,
     dst.copy_((src.float() * scale).to(dst.dtype))
 
 
+def emb_gather_sum(table: torch.Tensor, ids: torch.Tensor,
+                   out_bf16: bool = False) -> torch.Tensor:
+    """out[b] = sum_f table[ids[b, f], 0] for a scalar (dim-1) table."""
+    batch = ids.shape[0]
+    if _on_gpu(table, ids):
+        _require_ext()
+        return _C.emb_gather_sum(table, ids.reshape(-1).contiguous(),
+                                 batch, out_bf16)
+    out = table.reshape(-1).index_select(
+        0, ids.reshape(-1)).reshape(batch, -1).sum(dim=1)
+    return out.to(torch.bfloat16) if out_bf16 else out
+
+
+def emb_scatter_sum(table: torch.Tensor, ids: torch.Tensor,
+                    grad: torch.Tensor, alpha: float) -> None:
+    """table[ids[b, f], 0] += alpha * grad[b] (gather-sum backward)."""
+    if _on_gpu(table, ids, grad):
+        _require_ext()
+        _C.emb_scatter_sum(table, ids.reshape(-1).contiguous(),
+                           grad.contiguous(), alpha)
+        return
+    f = ids.reshape(grad.numel(), -1).shape[1]
+    expanded = grad.float().reshape(-1, 1).expand(-1, f).reshape(-1)
+    table.reshape(-1).index_add_(0, ids.reshape(-1), expanded,
+                                 alpha=alpha)
+
+
 class BiasReLU(torch.autograd.Function):
-    """Autograd wrapper for the fused bias+ReLU epilogue."""
+    """Autograd wrapper for the fused bias+ReLU epilogue (backward fuses
+    the dbias reduction into the dx kernel on GPU)."""
 
     @staticmethod
     def forward(ctx, x, bias):
@@ -198,6 +249,10 @@ class BiasReLU(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         (y,) = ctx.saved_tensors
+        if dy.is_cuda and HAVE_EXT and y.size(-1) <= 16384:
+            dx, dbias32 = _C.bias_relu_bwd_db(dy.contiguous(),
+                                              y.contiguous())
+            return dx, dbias32.to(dy.dtype)
         dx = bias_relu_bwd(dy.contiguous(), y)
         dims = tuple(range(dx.dim() - 1))
         dbias = dx.sum(dim=dims)

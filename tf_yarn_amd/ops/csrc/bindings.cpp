@@ -1,5 +1,6 @@
 // Python bindings for the tf_yarn_amd MI355X kernel library.
 #include <torch/extension.h>
+#include <vector>
 
 // fused_optimizers.hip
 void fused_sgd(torch::Tensor param, torch::Tensor grad,
@@ -8,6 +9,13 @@ void fused_sgd(torch::Tensor param, torch::Tensor grad,
                double lr, double momentum, double dampening,
                double weight_decay, bool nesterov, bool first_step,
                double grad_scale);
+void fused_sgd_mt(std::vector<torch::Tensor> params,
+                  std::vector<torch::Tensor> grads,
+                  std::vector<torch::Tensor> momentum_bufs,
+                  std::vector<torch::Tensor> params_bf16,
+                  double lr, double momentum, double dampening,
+                  double weight_decay, bool nesterov, bool first_step,
+                  double grad_scale);
 void fused_adam(torch::Tensor param, torch::Tensor grad,
                 torch::Tensor exp_avg, torch::Tensor exp_avg_sq,
                 c10::optional<torch::Tensor> param_bf16,
@@ -29,14 +37,23 @@ void emb_bwd_sgd(torch::Tensor table, torch::Tensor ids, torch::Tensor grad,
 void emb_bwd_dense(torch::Tensor grad_table, torch::Tensor ids,
                    torch::Tensor grad, double scale);
 
+torch::Tensor emb_gather_sum(torch::Tensor table, torch::Tensor ids,
+                             int64_t batch, bool out_bf16);
+void emb_scatter_sum(torch::Tensor table, torch::Tensor ids,
+                     torch::Tensor grad, double alpha);
+
 // elementwise.hip
 torch::Tensor bias_relu_fwd(torch::Tensor x, torch::Tensor bias);
 torch::Tensor bias_relu_bwd(torch::Tensor dy, torch::Tensor y);
+std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
+                                            torch::Tensor y);
 void convert_scaled(torch::Tensor src, torch::Tensor dst, double scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "tf_yarn_amd MI355X (gfx950) HIP kernels";
   m.def("fused_sgd", &fused_sgd, "Fused SGD(+momentum) step");
+  m.def("fused_sgd_mt", &fused_sgd_mt,
+        "Multi-tensor fused SGD (one launch per <=24 tensors)");
   m.def("fused_adam", &fused_adam, "Fused Adam/AdamW step");
   m.def("fused_adagrad", &fused_adagrad, "Fused Adagrad step");
   m.def("fused_adadelta", &fused_adadelta, "Fused Adadelta step");
@@ -47,5 +64,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Sparse embedding grad scatter into dense grad table");
   m.def("bias_relu_fwd", &bias_relu_fwd, "Fused bias+ReLU forward");
   m.def("bias_relu_bwd", &bias_relu_bwd, "Fused ReLU backward");
+  m.def("bias_relu_bwd_db", &bias_relu_bwd_db,
+        "Fused ReLU backward + dbias reduction (returns [dx, dbias_fp32])");
+  m.def("emb_gather_sum", &emb_gather_sum,
+        "Wide-part gather-sum: out[b] = sum_f table[ids[b,f]]");
+  m.def("emb_scatter_sum", &emb_scatter_sum,
+        "Wide-part scatter: table[ids[b,f]] += alpha * g[b]");
   m.def("convert_scaled", &convert_scaled, "Scaled bf16<->fp32 convert");
 }

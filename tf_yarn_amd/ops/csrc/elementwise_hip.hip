@@ -47,6 +47,34 @@ __global__ void bias_relu_bwd_kernel(
   }
 }
 
+// dx = dy * (y > 0) AND dbias[c] += sum_rows dx  (fused: the separate
+// torch bf16 reduce was 23 us x 3 layers per step in the profile).
+// Per-block LDS partials (fp32), one global atomicAdd per (block, col).
+template <typename Io>
+__global__ void bias_relu_bwd_db_kernel(
+    const typename Io::scalar_t* __restrict__ dy,
+    const typename Io::scalar_t* __restrict__ y,
+    typename Io::scalar_t* __restrict__ dx,
+    float* __restrict__ dbias,
+    int64_t rows, int64_t cols) {
+  extern __shared__ float sdb[];
+  for (int c = threadIdx.x; c < cols; c += blockDim.x) sdb[c] = 0.f;
+  __syncthreads();
+  const int64_t total = rows * cols;
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       i < total; i += stride) {
+    float g = Io::load(dy, i);
+    float yy = Io::load(y, i);
+    g = yy > 0.f ? g : 0.f;
+    Io::store(dx, i, g);
+    atomicAdd(&sdb[i % cols], g);
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < cols; c += blockDim.x)
+    if (sdb[c] != 0.f) atomicAdd(&dbias[c], sdb[c]);
+}
+
 __global__ void bf16_to_f32_kernel(const unsigned short* __restrict__ src,
                                    float* __restrict__ dst, int64_t n,
                                    float scale) {
@@ -133,6 +161,38 @@ torch::Tensor bias_relu_bwd(torch::Tensor dy, torch::Tensor y) {
                        total);
   }
   return dx;
+}
+
+std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
+                                            torch::Tensor y) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && y.is_contiguous(),
+              "dy/y must be GPU contiguous");
+  TORCH_CHECK(dy.scalar_type() == y.scalar_type() &&
+              dy.numel() == y.numel(), "dy/y mismatch");
+  const int64_t cols = dy.size(-1);
+  const int64_t rows = dy.numel() / cols;
+  TORCH_CHECK(cols <= 16384, "bias_relu_bwd_db supports cols <= 16384");
+  auto dx = torch::empty_like(dy);
+  auto dbias = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int grid = miyarn_grid(rows * cols);
+  size_t lds = cols * sizeof(float);
+  if (dy.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(bias_relu_bwd_db_kernel<F32Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), lds, stream,
+                       dy.data_ptr<float>(), y.data_ptr<float>(),
+                       dx.data_ptr<float>(), dbias.data_ptr<float>(),
+                       rows, cols);
+  } else {
+    TORCH_CHECK(dy.scalar_type() == torch::kBFloat16, "fp32/bf16 only");
+    hipLaunchKernelGGL(bias_relu_bwd_db_kernel<Bf16Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), lds, stream,
+                       reinterpret_cast<unsigned short*>(dy.data_ptr()),
+                       reinterpret_cast<unsigned short*>(y.data_ptr()),
+                       reinterpret_cast<unsigned short*>(dx.data_ptr()),
+                       dbias.data_ptr<float>(), rows, cols);
+  }
+  return {dx, dbias};
 }
 
 void convert_scaled(torch::Tensor src, torch::Tensor dst, double scale) {

@@ -119,6 +119,39 @@ __global__ void emb_scatter_scalar_kernel(
   }
 }
 
+// Wide-part fusion: out[b] = sum_f table[ids[b*F + f]] for scalar
+// (dim-1) tables — replaces gather + torch reduce with one kernel.
+template <typename OIo>
+__global__ void emb_gather_sum_kernel(const float* __restrict__ table,
+                                      const int64_t* __restrict__ ids,
+                                      typename OIo::scalar_t* __restrict__ out,
+                                      int64_t batch, int F) {
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t b = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       b < batch; b += stride) {
+    float acc = 0.f;
+    const int64_t* row = ids + b * F;
+#pragma unroll 4
+    for (int f = 0; f < F; ++f) acc += table[row[f]];
+    OIo::store(out, b, acc);
+  }
+}
+
+// Backward of gather-sum: table[ids[b*F+f]] += alpha * g[b]
+template <typename GIo>
+__global__ void emb_scatter_sum_kernel(float* __restrict__ table,
+                                       const int64_t* __restrict__ ids,
+                                       const typename GIo::scalar_t* __restrict__ g,
+                                       int64_t batch, int F, float alpha) {
+  const int64_t total = batch * (int64_t)F;
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       t < total; t += stride) {
+    const int64_t b = t / F;
+    atomicAdd(table + ids[t], alpha * GIo::load(g, b));
+  }
+}
+
 void check_emb(const torch::Tensor& table, const torch::Tensor& ids,
                int64_t dim) {
   TORCH_CHECK(table.is_cuda() && table.is_contiguous() &&
@@ -247,5 +280,61 @@ void emb_bwd_dense(torch::Tensor grad_table, torch::Tensor ids,
                          grad_table.data_ptr<float>(),
                          ids.data_ptr<int64_t>(), gptr, n, dim,
                          (float)scale);
+  }
+}
+
+torch::Tensor emb_gather_sum(torch::Tensor table, torch::Tensor ids,
+                             int64_t batch, bool out_bf16) {
+  TORCH_CHECK(table.is_cuda() && table.is_contiguous() &&
+              table.scalar_type() == torch::kFloat32,
+              "table must be contiguous fp32 on GPU");
+  TORCH_CHECK(ids.is_cuda() && ids.is_contiguous() &&
+              ids.scalar_type() == torch::kInt64, "ids must be int64 GPU");
+  TORCH_CHECK(ids.numel() % batch == 0, "ids size not divisible by batch");
+  const int F = static_cast<int>(ids.numel() / batch);
+  auto out = torch::empty({batch}, table.options().dtype(
+      out_bf16 ? torch::kBFloat16 : torch::kFloat32));
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int grid = miyarn_grid(batch);
+  if (out_bf16) {
+    hipLaunchKernelGGL(emb_gather_sum_kernel<Bf16Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream,
+                       table.data_ptr<float>(), ids.data_ptr<int64_t>(),
+                       reinterpret_cast<unsigned short*>(out.data_ptr()),
+                       batch, F);
+  } else {
+    hipLaunchKernelGGL(emb_gather_sum_kernel<F32Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream,
+                       table.data_ptr<float>(), ids.data_ptr<int64_t>(),
+                       out.data_ptr<float>(), batch, F);
+  }
+  return out;
+}
+
+void emb_scatter_sum(torch::Tensor table, torch::Tensor ids,
+                     torch::Tensor grad, double alpha) {
+  TORCH_CHECK(table.is_cuda() && table.is_contiguous() &&
+              table.scalar_type() == torch::kFloat32,
+              "table must be contiguous fp32 on GPU");
+  TORCH_CHECK(ids.is_cuda() && ids.is_contiguous() &&
+              ids.scalar_type() == torch::kInt64, "ids must be int64 GPU");
+  const int64_t batch = grad.numel();
+  TORCH_CHECK(ids.numel() % batch == 0, "ids size not divisible by batch");
+  const int F = static_cast<int>(ids.numel() / batch);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int grid = miyarn_grid(batch * F);
+  if (grad.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(emb_scatter_sum_kernel<F32Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream,
+                       table.data_ptr<float>(), ids.data_ptr<int64_t>(),
+                       grad.data_ptr<float>(), batch, F, (float)alpha);
+  } else {
+    TORCH_CHECK(grad.scalar_type() == torch::kBFloat16,
+                "grad must be fp32 or bf16");
+    hipLaunchKernelGGL(emb_scatter_sum_kernel<Bf16Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream,
+                       table.data_ptr<float>(), ids.data_ptr<int64_t>(),
+                       reinterpret_cast<unsigned short*>(grad.data_ptr()),
+                       batch, F, (float)alpha);
   }
 }

@@ -193,6 +193,77 @@ __global__ void adadelta_kernel(float* __restrict__ p,
   }
 }
 
+// ---- multi-tensor SGD ------------------------------------------------------
+// One launch for all dense params of a group (the per-param launches were
+// ~10 x 4.6 us/step in the profile): gridDim.y = tensor index, each y-slice
+// grid-strides its own tensor.  Kernel-arg struct holds up to MT_MAX
+// tensors (HIP kernarg limit 4 KB).
+
+#define MT_MAX 24
+
+struct MtTensors {
+  float* p[MT_MAX];
+  const void* g[MT_MAX];
+  float* m[MT_MAX];           // nullptr if no momentum
+  unsigned short* pb[MT_MAX]; // nullptr if no bf16 copy
+  long numel[MT_MAX];
+  int n;
+};
+
+template <typename GIo>
+__global__ void sgd_mt_kernel(MtTensors t, SgdArgs a) {
+  const int ti = blockIdx.y;
+  if (ti >= t.n) return;
+  const int64_t n = t.numel[ti];
+  float* __restrict__ p = t.p[ti];
+  const typename GIo::scalar_t* __restrict__ g =
+      reinterpret_cast<const typename GIo::scalar_t*>(t.g[ti]);
+  float* __restrict__ mom = t.m[ti];
+  unsigned short* __restrict__ pb = t.pb[ti];
+  const int64_t tid = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  const int64_t nvec = n >> 2;
+  for (int64_t i = tid; i < nvec; i += stride) {
+    f32x4 pv = reinterpret_cast<f32x4*>(p)[i];
+    f32x4 mv;
+    if (mom) mv = reinterpret_cast<f32x4*>(mom)[i];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float gv = GIo::load(g, i * 4 + j) * a.scale;
+      gv += a.weight_decay * pv[j];
+      if (mom) {
+        float m = a.first_step ? gv
+                               : a.momentum * mv[j] + (1.f - a.dampening) * gv;
+        mv[j] = m;
+        gv = a.nesterov ? gv + a.momentum * m : m;
+      }
+      pv[j] -= a.lr * gv;
+    }
+    reinterpret_cast<f32x4*>(p)[i] = pv;
+    if (mom) reinterpret_cast<f32x4*>(mom)[i] = mv;
+    if (pb) {
+      bf16x4 bv;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) bv[j] = f32_to_bf16(pv[j]);
+      reinterpret_cast<bf16x4*>(pb)[i] = bv;
+    }
+  }
+  for (int64_t i = (nvec << 2) + tid; i < n; i += stride) {
+    float pv = p[i];
+    float gv = GIo::load(g, i) * a.scale;
+    gv += a.weight_decay * pv;
+    if (mom) {
+      float m = a.first_step ? gv
+                             : a.momentum * mom[i] + (1.f - a.dampening) * gv;
+      mom[i] = m;
+      gv = a.nesterov ? gv + a.momentum * m : m;
+    }
+    pv -= a.lr * gv;
+    p[i] = pv;
+    if (pb) pb[i] = f32_to_bf16(pv);
+  }
+}
+
 // ---- host-side checks/dispatch --------------------------------------------
 
 void check_flat_f32(const torch::Tensor& t, const char* name, int64_t n) {
@@ -249,6 +320,62 @@ void fused_sgd(torch::Tensor param, torch::Tensor grad,
                        stream, param.data_ptr<float>(),
                        reinterpret_cast<unsigned short*>(grad.data_ptr()),
                        mom, bf16_copy_ptr(param_bf16, n), n, a);
+  }
+}
+
+void fused_sgd_mt(std::vector<torch::Tensor> params,
+                  std::vector<torch::Tensor> grads,
+                  std::vector<torch::Tensor> momentum_bufs,  // may be empty
+                  std::vector<torch::Tensor> params_bf16,    // may be empty
+                  double lr, double momentum, double dampening,
+                  double weight_decay, bool nesterov, bool first_step,
+                  double grad_scale) {
+  const int n = static_cast<int>(params.size());
+  TORCH_CHECK(n > 0 && (int)grads.size() == n, "params/grads mismatch");
+  const bool has_mom = !momentum_bufs.empty();
+  const bool has_bf16 = !params_bf16.empty();
+  TORCH_CHECK(!has_mom || (int)momentum_bufs.size() == n, "momentum size");
+  TORCH_CHECK(!has_bf16 || (int)params_bf16.size() == n, "bf16 size");
+  auto gtype = grads[0].scalar_type();
+  SgdArgs a{(float)lr, (float)momentum, (float)dampening,
+            (float)weight_decay, (float)grad_scale, nesterov, first_step};
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  for (int base = 0; base < n; base += MT_MAX) {
+    MtTensors t{};
+    int k = std::min(MT_MAX, n - base);
+    t.n = k;
+    int64_t max_numel = 0;
+    for (int i = 0; i < k; ++i) {
+      const auto& p = params[base + i];
+      const auto& g = grads[base + i];
+      int64_t ne = p.numel();
+      check_flat_f32(p, "param", ne);
+      check_grad(g, ne);
+      TORCH_CHECK(g.scalar_type() == gtype,
+                  "all grads in one fused_sgd_mt call share a dtype");
+      t.p[i] = p.data_ptr<float>();
+      t.g[i] = g.data_ptr();
+      t.m[i] = nullptr;
+      t.pb[i] = nullptr;
+      if (has_mom) {
+        check_flat_f32(momentum_bufs[base + i], "momentum", ne);
+        t.m[i] = momentum_bufs[base + i].data_ptr<float>();
+      }
+      if (has_bf16) {
+        t.pb[i] = bf16_copy_ptr(
+            c10::optional<torch::Tensor>(params_bf16[base + i]), ne);
+      }
+      t.numel[i] = ne;
+      max_numel = std::max(max_numel, ne);
+    }
+    dim3 grid(miyarn_grid((max_numel + 3) / 4), k);
+    if (gtype == torch::kFloat32) {
+      hipLaunchKernelGGL(sgd_mt_kernel<F32Io>, grid, dim3(MIYARN_BLOCK), 0,
+                         stream, t, a);
+    } else {
+      hipLaunchKernelGGL(sgd_mt_kernel<Bf16Io>, grid, dim3(MIYARN_BLOCK), 0,
+                         stream, t, a);
+    }
   }
 }
 

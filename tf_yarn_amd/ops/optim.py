@@ -46,6 +46,9 @@ class FusedSGD(Optimizer):
     def step(self, closure=None):
         loss = closure() if closure is not None else None
         for group in self.param_groups:
+            # Bucket params by (grad dtype, momentum?, bf16 copy?, first?)
+            # so each bucket is one multi-tensor kernel launch.
+            buckets = {}
             for p in group["params"]:
                 if getattr(p, "_miyarn_sparse", False):
                     continue
@@ -61,17 +64,28 @@ class FusedSGD(Optimizer):
                     if "momentum_buffer" not in state:
                         state["momentum_buffer"] = torch.zeros_like(master)
                     mom_buf = state["momentum_buffer"]
-                ops.fused_sgd(
-                    master.view(-1), g.view(-1),
-                    mom_buf.view(-1) if mom_buf is not None else None,
-                    bf16.view(-1) if bf16 is not None else None,
+                key = (g.dtype, mom_buf is not None, bf16 is not None,
+                       first)
+                b = buckets.setdefault(key, ([], [], [], []))
+                b[0].append(master.view(-1))
+                b[1].append(g.view(-1))
+                if mom_buf is not None:
+                    b[2].append(mom_buf.view(-1))
+                if bf16 is not None:
+                    b[3].append(bf16.view(-1))
+                if bf16 is None and p.dtype != torch.float32:
+                    p.data.copy_(master.to(p.dtype))
+            for (gdtype, has_mom, has_bf16, first), \
+                    (masters, grads, moms, bf16s) in buckets.items():
+                ops.fused_sgd_mt(
+                    masters, grads,
+                    moms if has_mom else None,
+                    bf16s if has_bf16 else None,
                     lr=group["lr"], momentum=group["momentum"],
                     dampening=group["dampening"],
                     weight_decay=group["weight_decay"],
                     nesterov=group["nesterov"], first_step=first,
                     grad_scale=group["grad_scale"])
-                if bf16 is None and p.dtype != torch.float32:
-                    p.data.copy_(master.to(p.dtype))
         return loss
 
 
