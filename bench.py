@@ -21,6 +21,12 @@ import os
 import sys
 import time
 
+# hipBLASLt algorithm tuning for the MLP GEMM shapes (must be set before
+# the first GEMM; tuning happens during the untimed warmup steps).
+os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+os.environ.setdefault("PYTORCH_TUNABLEOP_VERBOSE", "0")
+
 import torch
 import torch.distributed as dist
 

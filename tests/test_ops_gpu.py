@@ -229,8 +229,8 @@ def test_bias_relu_bwd_db_matches_reference():
     import tf_yarn_amd.ops._C as C
     torch.manual_seed(10)
     for dtype in (torch.float32, torch.bfloat16):
-        y = torch.relu(torch.randn(1024, 513)).cuda().to(dtype)
-        dy = torch.randn(1024, 513).cuda().to(dtype)
+        y = torch.relu(torch.randn(1024, 516)).cuda().to(dtype)
+        dy = torch.randn(1024, 516).cuda().to(dtype)
         dx, dbias = C.bias_relu_bwd_db(dy.contiguous(), y.contiguous())
         ref_dx = dy.float() * (y.float() > 0)
         assert torch.allclose(dx.float(), ref_dx.to(dtype).float(),

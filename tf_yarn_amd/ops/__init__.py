@@ -249,7 +249,8 @@ class BiasReLU(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         (y,) = ctx.saved_tensors
-        if dy.is_cuda and HAVE_EXT and y.size(-1) <= 16384:
+        if dy.is_cuda and HAVE_EXT and y.size(-1) % 4 == 0 \
+                and y.size(-1) <= 8192:
             dx, dbias32 = _C.bias_relu_bwd_db(dy.contiguous(),
                                               y.contiguous())
             return dx, dbias32.to(dy.dtype)

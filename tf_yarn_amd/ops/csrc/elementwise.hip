@@ -48,7 +48,12 @@ __global__ void bias_relu_bwd_kernel(
 
 // dx = dy * (y > 0) AND dbias[c] += sum_rows dx  (fused: the separate
 // torch bf16 reduce was 23 us x 3 layers per step in the profile).
-// Per-block LDS partials (fp32), one global atomicAdd per (block, col).
+// Row-wise: each thread owns 4 consecutive columns (8 B/lane bf16 loads,
+// coalesced), accumulates its column partials in registers over its
+// block's rows, then does ONE atomicAdd per owned column.  Column count
+// per thread = cols / (4 * blockDim) rounded up, capped at DB_MAX_K.
+#define DB_MAX_K 8
+
 template <typename Io>
 __global__ void bias_relu_bwd_db_kernel(
     const typename Io::scalar_t* __restrict__ dy,
@@ -56,22 +61,33 @@ __global__ void bias_relu_bwd_db_kernel(
     typename Io::scalar_t* __restrict__ dx,
     float* __restrict__ dbias,
     int64_t rows, int64_t cols) {
-  extern __shared__ float sdb[];
-  for (int c = threadIdx.x; c < cols; c += blockDim.x) sdb[c] = 0.f;
-  __syncthreads();
-  const int64_t total = rows * cols;
-  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
-  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
-       i < total; i += stride) {
-    float g = Io::load(dy, i);
-    float yy = Io::load(y, i);
-    g = yy > 0.f ? g : 0.f;
-    Io::store(dx, i, g);
-    atomicAdd(&sdb[i % cols], g);
+  const int64_t quads = cols >> 2;  // cols % 4 == 0 guaranteed by host
+  float acc[DB_MAX_K][4];
+#pragma unroll
+  for (int k = 0; k < DB_MAX_K; ++k)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[k][j] = 0.f;
+  for (int64_t r = blockIdx.x; r < rows; r += gridDim.x) {
+    const int64_t base = r * cols;
+    int k = 0;
+    for (int64_t q = threadIdx.x; q < quads; q += blockDim.x, ++k) {
+      const int64_t i = base + q * 4;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float g = Io::load(dy, i + j);
+        float yy = Io::load(y, i + j);
+        g = yy > 0.f ? g : 0.f;
+        Io::store(dx, i + j, g);
+        acc[k][j] += g;
+      }
+    }
   }
-  __syncthreads();
-  for (int c = threadIdx.x; c < cols; c += blockDim.x)
-    if (sdb[c] != 0.f) atomicAdd(&dbias[c], sdb[c]);
+  int k = 0;
+  for (int64_t q = threadIdx.x; q < quads; q += blockDim.x, ++k) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      atomicAdd(&dbias[q * 4 + j], acc[k][j]);
+  }
 }
 
 __global__ void bf16_to_f32_kernel(const unsigned short* __restrict__ src,
@@ -170,12 +186,14 @@ std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
               dy.numel() == y.numel(), "dy/y mismatch");
   const int64_t cols = dy.size(-1);
   const int64_t rows = dy.numel() / cols;
-  TORCH_CHECK(cols <= 16384, "bias_relu_bwd_db supports cols <= 16384");
+  TORCH_CHECK(cols % 4 == 0 && cols <= 4 * MIYARN_BLOCK * DB_MAX_K,
+              "bias_relu_bwd_db needs cols % 4 == 0 and cols <= ",
+              4 * MIYARN_BLOCK * DB_MAX_K);
   auto dx = torch::empty_like(dy);
   auto dbias = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  int grid = miyarn_grid(rows * cols);
-  size_t lds = cols * sizeof(float);
+  int grid = static_cast<int>(std::min<int64_t>(rows, MIYARN_MAX_BLOCKS));
+  size_t lds = 0;
   if (dy.scalar_type() == torch::kFloat32) {
     hipLaunchKernelGGL(bias_relu_bwd_db_kernel<F32Io>, dim3(grid),
                        dim3(MIYARN_BLOCK), lds, stream,
