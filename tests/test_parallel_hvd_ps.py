@@ -1,0 +1,110 @@
+"""Horovod-style optimizer + PS engine tests (CPU, gloo where needed)."""
+
+import threading
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+from torch import nn
+
+from tf_yarn_amd.kv import KVClient, KVServer
+from tf_yarn_amd.parallel.ps import PsTopology, _ShardLayout, \
+    shard_parameters
+from tf_yarn_amd.topologies import ContainerTask
+
+
+def test_shard_parameters_balanced():
+    shapes = [(100, 10), (50,), (200, 5), (3,)]
+    shards = shard_parameters(shapes, 2)
+    assert sorted(i for s in shards for i in s) == [0, 1, 2, 3]
+    # the two big tensors land on different shards
+    big = {0, 2}
+    assert not big.issubset(set(shards[0]))
+    assert not big.issubset(set(shards[1]))
+
+
+def test_shard_layout_pack_unpack_roundtrip():
+    params = [torch.randn(10, 4), torch.randn(7), torch.randn(3, 3)]
+    layout = _ShardLayout(params, 2)
+    outs = [torch.zeros(n) for n in layout.shard_numel]
+    for k in range(2):
+        layout.pack(k, params, outs[k])
+    restored = [torch.zeros_like(p) for p in params]
+    for k in range(2):
+        layout.unpack(k, outs[k], restored)
+    for p, r in zip(params, restored):
+        assert torch.equal(p, r)
+
+
+def test_ps_topology_ranks():
+    tasks = [ContainerTask("chief", 0, 1), ContainerTask("ps", 0, 1),
+             ContainerTask("ps", 1, 1), ContainerTask("worker", 0, 1),
+             ContainerTask("worker", 1, 1)]
+    topo_chief = PsTopology(tasks, "chief", 0)
+    assert topo_chief.rank == 0 and not topo_chief.is_ps
+    assert topo_chief.n_workers == 3 and topo_chief.n_ps == 2
+    topo_w1 = PsTopology(tasks, "worker", 1)
+    assert topo_w1.rank == 2
+    topo_ps1 = PsTopology(tasks, "ps", 1)
+    assert topo_ps1.rank == 4 and topo_ps1.is_ps
+    assert topo_chief.ps_ranks == [3, 4]
+    with pytest.raises(ValueError):
+        PsTopology(tasks, "worker", 9)
+
+
+def _hvd_worker(rank, world_size, kv_addr, out_q):
+    from tf_yarn_amd.parallel import comm
+    from tf_yarn_amd.parallel.hvd import (DistributedOptimizer,
+                                          broadcast_parameters)
+    client = KVClient(kv_addr)
+    comm.init_process_group(rank=rank, world_size=world_size,
+                            backend="gloo", kv_client=client)
+    try:
+        torch.manual_seed(rank * 7 + 1)  # deliberately different init
+        model = nn.Linear(4, 2)
+        broadcast_parameters(model, root_rank=0)
+        opt = DistributedOptimizer(
+            torch.optim.SGD(model.parameters(), lr=0.1))
+        for step in range(3):
+            torch.manual_seed(50 + step * world_size + rank)
+            x = torch.randn(8, 4)
+            y = torch.randn(8, 2)
+            opt.zero_grad()
+            nn.functional.mse_loss(model(x), y).backward()
+            opt.step()
+        out_q.put((rank, [p.detach().numpy().copy()
+                          for p in model.parameters()]))
+    finally:
+        comm.destroy_process_group()
+
+
+def test_hvd_distributed_optimizer_converges_identically():
+    """After rank-0 broadcast + allreduce steps, all ranks hold identical
+    params (the Horovod-path contract)."""
+    world_size = 2
+    server = KVServer()
+    ctx = mp.get_context("spawn")
+    out_q = ctx.Queue()
+    procs = [ctx.Process(target=_hvd_worker,
+                         args=(r, world_size, server.address, out_q))
+             for r in range(world_size)]
+    for p in procs:
+        p.start()
+    try:
+        results = dict(out_q.get(timeout=120) for _ in range(world_size))
+    finally:
+        for p in procs:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.terminate()
+        server.stop()
+    for a, b in zip(results[0], results[1]):
+        assert (a == b).all(), "ranks diverged"
+
+
+def test_allreduce_tensors_single_process_noop():
+    from tf_yarn_amd.parallel.hvd import allreduce_tensors
+    t = torch.randn(5)
+    ref = t.clone()
+    allreduce_tensors([t])
+    assert torch.equal(t, ref)
