@@ -1,0 +1,121 @@
+#!/usr/bin/env python3
+"""Secondary benchmark: ResNet-50 DDP bf16 on synthetic ImageNet
+(BASELINE config 4).  Same JSON contract and launch protocol as bench.py.
+
+    python scripts/bench_resnet.py --gpus N --steps K --warmup W
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+import torch.distributed as dist
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from tf_yarn_amd.models.resnet import resnet50  # noqa: E402
+from tf_yarn_amd.ops.optim import FusedSGD  # noqa: E402
+from tf_yarn_amd.parallel.ddp import BucketedDataParallel  # noqa: E402
+
+PER_GPU_BATCH = 256
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=PER_GPU_BATCH)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world_size = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    use_gpu = torch.cuda.is_available()
+    device = f"cuda:{local_rank}" if use_gpu else "cpu"
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+    if world_size > 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29500")
+        dist.init_process_group("nccl" if use_gpu else "gloo",
+                                rank=rank, world_size=world_size)
+
+    torch.manual_seed(42)
+    torch.backends.cudnn.benchmark = True
+    model = resnet50().to(device)
+    dtype = torch.bfloat16 if use_gpu else torch.float32
+    model = model.to(dtype)
+    # BN stays in fp32 for numerics
+    for m in model.modules():
+        if isinstance(m, torch.nn.BatchNorm2d):
+            m.float()
+    ddp = BucketedDataParallel(model, broadcast_buffers=False) \
+        if world_size > 1 else model
+    opt = FusedSGD(model.parameters(), lr=0.1, momentum=0.9,
+                   weight_decay=1e-4)
+    loss_fn = torch.nn.CrossEntropyLoss()
+
+    x = torch.randn(args.batch, 3, 224, 224, device=device, dtype=dtype)
+    y = torch.randint(0, 1000, (args.batch,), device=device)
+
+    def step():
+        opt.zero_grad(set_to_none=False)
+        out = ddp(x)
+        loss = loss_fn(out.float(), y)
+        loss.backward()
+        opt.step()
+        return loss
+
+    for _ in range(args.warmup):
+        step()
+    if use_gpu:
+        torch.cuda.synchronize()
+    if world_size > 1:
+        dist.barrier()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        loss = step()
+    if use_gpu:
+        torch.cuda.synchronize()
+    if world_size > 1:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if use_gpu else "cpu")
+    if world_size > 1:
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "images/sec (whole node), ResNet-50 DDP",
+            "value": args.batch * world_size * args.steps / elapsed,
+            "unit": "images/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_gpu else "fp32",
+            "data": "synthetic",
+            "config": {"model": "resnet50",
+                       "global_batch": args.batch * world_size,
+                       "per_gpu_batch": args.batch,
+                       "image": [3, 224, 224],
+                       "parallelism": f"dp{world_size}",
+                       "final_loss": float(loss.item())},
+        }), flush=True)
+    if world_size > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -77,3 +77,17 @@ def test_bf16_compute_mode_cpu_fallback():
         logits.float(), labels)
     loss.backward()
     model.apply_sparse_updates(lr=0.1)
+
+
+def test_resnet_forward_backward_cpu():
+    from tf_yarn_amd.models.resnet import resnet18_like, resnet50
+    import torch
+    from torch import nn
+    m = resnet18_like(num_classes=10)
+    x = torch.randn(2, 3, 64, 64)
+    out = m(x)
+    assert out.shape == (2, 10)
+    nn.functional.cross_entropy(out, torch.tensor([1, 2])).backward()
+    # resnet50 has the canonical parameter count (~25.6M)
+    n_params = sum(p.numel() for p in resnet50().parameters())
+    assert abs(n_params - 25_557_032) < 1000, n_params

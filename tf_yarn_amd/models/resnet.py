@@ -1,0 +1,96 @@
+"""ResNet-50 for the synthetic-ImageNet DDP benchmark (BASELINE config 4).
+
+Hand-written (torchvision is not in this environment): the standard
+bottleneck-v1 architecture with torch conv/bn ops — convolutions go
+through MIOpen, the dense data-parallel path through the framework's
+bucketed-allreduce reducer.
+"""
+
+from __future__ import annotations
+
+from typing import List, Type
+
+import torch
+from torch import nn
+
+
+class Bottleneck(nn.Module):
+    expansion = 4
+
+    def __init__(self, in_planes: int, planes: int, stride: int = 1,
+                 downsample: nn.Module = None):
+        super().__init__()
+        width = planes
+        self.conv1 = nn.Conv2d(in_planes, width, 1, bias=False)
+        self.bn1 = nn.BatchNorm2d(width)
+        self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
+                               bias=False)
+        self.bn2 = nn.BatchNorm2d(width)
+        self.conv3 = nn.Conv2d(width, planes * self.expansion, 1,
+                               bias=False)
+        self.bn3 = nn.BatchNorm2d(planes * self.expansion)
+        self.relu = nn.ReLU(inplace=True)
+        self.downsample = downsample
+        self.stride = stride
+
+    def forward(self, x):
+        identity = x
+        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.relu(self.bn2(self.conv2(out)))
+        out = self.bn3(self.conv3(out))
+        if self.downsample is not None:
+            identity = self.downsample(x)
+        return self.relu(out + identity)
+
+
+class ResNet(nn.Module):
+    def __init__(self, layers: List[int], num_classes: int = 1000):
+        super().__init__()
+        self.in_planes = 64
+        self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
+        self.bn1 = nn.BatchNorm2d(64)
+        self.relu = nn.ReLU(inplace=True)
+        self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+        self.layer1 = self._make_layer(64, layers[0])
+        self.layer2 = self._make_layer(128, layers[1], stride=2)
+        self.layer3 = self._make_layer(256, layers[2], stride=2)
+        self.layer4 = self._make_layer(512, layers[3], stride=2)
+        self.avgpool = nn.AdaptiveAvgPool2d(1)
+        self.fc = nn.Linear(512 * Bottleneck.expansion, num_classes)
+        for m in self.modules():
+            if isinstance(m, nn.Conv2d):
+                nn.init.kaiming_normal_(m.weight, mode="fan_out",
+                                        nonlinearity="relu")
+            elif isinstance(m, nn.BatchNorm2d):
+                nn.init.constant_(m.weight, 1)
+                nn.init.constant_(m.bias, 0)
+
+    def _make_layer(self, planes: int, blocks: int,
+                    stride: int = 1) -> nn.Sequential:
+        downsample = None
+        out_planes = planes * Bottleneck.expansion
+        if stride != 1 or self.in_planes != out_planes:
+            downsample = nn.Sequential(
+                nn.Conv2d(self.in_planes, out_planes, 1, stride=stride,
+                          bias=False),
+                nn.BatchNorm2d(out_planes))
+        layers = [Bottleneck(self.in_planes, planes, stride, downsample)]
+        self.in_planes = out_planes
+        for _ in range(1, blocks):
+            layers.append(Bottleneck(self.in_planes, planes))
+        return nn.Sequential(*layers)
+
+    def forward(self, x):
+        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
+        x = self.avgpool(x).flatten(1)
+        return self.fc(x)
+
+
+def resnet50(num_classes: int = 1000) -> ResNet:
+    return ResNet([3, 4, 6, 3], num_classes)
+
+
+def resnet18_like(num_classes: int = 1000) -> ResNet:
+    """Small variant for tests."""
+    return ResNet([1, 1, 1, 1], num_classes)
