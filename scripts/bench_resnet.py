@@ -13,6 +13,10 @@ import os
 import sys
 import time
 
+# MIOpen exhaustive per-conv search takes minutes on a fresh box; FAST
+# find keeps warmup within the bench budget.
+os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+
 import torch
 import torch.distributed as dist
 
