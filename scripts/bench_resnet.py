@@ -35,6 +35,7 @@ def main() -> None:
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--batch", type=int, default=PER_GPU_BATCH)
+    ap.add_argument("--channels-last", action="store_true")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -66,6 +67,9 @@ def main() -> None:
     loss_fn = torch.nn.CrossEntropyLoss()
 
     x = torch.randn(args.batch, 3, 224, 224, device=device, dtype=dtype)
+    if args.channels_last:
+        model = model.to(memory_format=torch.channels_last)
+        x = x.to(memory_format=torch.channels_last)
     y = torch.randint(0, 1000, (args.batch,), device=device)
 
     def step():

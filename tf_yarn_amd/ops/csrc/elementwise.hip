@@ -54,6 +54,8 @@ __global__ void bias_relu_bwd_kernel(
 // per thread = cols / (4 * blockDim) rounded up, capped at DB_MAX_K.
 #define DB_MAX_K 8
 
+#define DB_ROWS 4  // rows in flight per thread (ILP for the HBM latency)
+
 template <typename Io>
 __global__ void bias_relu_bwd_db_kernel(
     const typename Io::scalar_t* __restrict__ dy,
@@ -67,19 +69,32 @@ __global__ void bias_relu_bwd_db_kernel(
   for (int k = 0; k < DB_MAX_K; ++k)
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[k][j] = 0.f;
-  for (int64_t r = blockIdx.x; r < rows; r += gridDim.x) {
-    const int64_t base = r * cols;
+  const int64_t row_stride = (int64_t)gridDim.x * DB_ROWS;
+  for (int64_t r0 = (int64_t)blockIdx.x * DB_ROWS; r0 < rows;
+       r0 += row_stride) {
+    const int nr = min((int64_t)DB_ROWS, rows - r0);
     int k = 0;
     for (int64_t q = threadIdx.x; q < quads; q += blockDim.x, ++k) {
-      const int64_t i = base + q * 4;
+      float v[DB_ROWS][4];
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        float g = Io::load(dy, i + j);
-        float yy = Io::load(y, i + j);
-        g = yy > 0.f ? g : 0.f;
-        Io::store(dx, i + j, g);
-        acc[k][j] += g;
-      }
+      for (int rr = 0; rr < DB_ROWS; ++rr)
+        if (rr < nr)
+          QuadIo<Io>::load4(dy, (r0 + rr) * quads + q, v[rr]);
+      float yy[DB_ROWS][4];
+#pragma unroll
+      for (int rr = 0; rr < DB_ROWS; ++rr)
+        if (rr < nr)
+          QuadIo<Io>::load4(y, (r0 + rr) * quads + q, yy[rr]);
+#pragma unroll
+      for (int rr = 0; rr < DB_ROWS; ++rr)
+        if (rr < nr) {
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            v[rr][j] = yy[rr][j] > 0.f ? v[rr][j] : 0.f;
+            acc[k][j] += v[rr][j];
+          }
+          QuadIo<Io>::store4(dx, (r0 + rr) * quads + q, v[rr]);
+        }
     }
   }
   int k = 0;
@@ -192,7 +207,8 @@ std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
   auto dx = torch::empty_like(dy);
   auto dbias = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  int grid = static_cast<int>(std::min<int64_t>(rows, MIYARN_MAX_BLOCKS));
+  int grid = static_cast<int>(std::min<int64_t>(
+      (rows + DB_ROWS - 1) / DB_ROWS, MIYARN_MAX_BLOCKS));
   size_t lds = 0;
   if (dy.scalar_type() == torch::kFloat32) {
     hipLaunchKernelGGL(bias_relu_bwd_db_kernel<F32Io>, dim3(grid),

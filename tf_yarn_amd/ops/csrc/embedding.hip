@@ -38,9 +38,10 @@ __global__ void emb_fwd_kernel(const float* __restrict__ table,
     const int64_t c4 = t - row * dvec;
     const int64_t src = ids[row];
     f32x4 v = reinterpret_cast<const f32x4*>(table + src * dim)[c4];
+    float vv[4];
 #pragma unroll
-    for (int j = 0; j < 4; ++j)
-      OIo::store(out, row * dim + c4 * 4 + j, v[j]);
+    for (int j = 0; j < 4; ++j) vv[j] = v[j];
+    QuadIo<OIo>::store4(out, row * dvec + c4, vv);
   }
 }
 
@@ -58,11 +59,11 @@ __global__ void emb_bwd_sgd_kernel(float* __restrict__ table,
     const int64_t row = t / dvec;
     const int64_t c4 = t - row * dvec;
     float* dst = table + ids[row] * dim + c4 * 4;
+    float gv[4];
+    QuadIo<GIo>::load4(g, row * dvec + c4, gv);
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      float gv = GIo::load(g, row * dim + c4 * 4 + j);
-      atomicAdd(dst + j, neg_lr_scale * gv);
-    }
+    for (int j = 0; j < 4; ++j)
+      atomicAdd(dst + j, neg_lr_scale * gv[j]);
   }
 }
 
@@ -80,11 +81,11 @@ __global__ void emb_bwd_dense_kernel(
     const int64_t row = t / dvec;
     const int64_t c4 = t - row * dvec;
     float* dst = grad_table + ids[row] * dim + c4 * 4;
+    float gv[4];
+    QuadIo<GIo>::load4(g, row * dvec + c4, gv);
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      float gv = GIo::load(g, row * dim + c4 * 4 + j);
-      atomicAdd(dst + j, scale * gv);
-    }
+    for (int j = 0; j < 4; ++j)
+      atomicAdd(dst + j, scale * gv[j]);
   }
 }
 
