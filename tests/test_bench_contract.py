@@ -1,0 +1,59 @@
+"""Guard the driver's bench.py contract: single JSON line on stdout with
+the required fields, both single-process and under torch.distributed.run
+(CPU/gloo here; the driver runs the same launch shape on MI355X)."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+REQUIRED_FIELDS = [
+    "metric", "value", "unit", "n_gpus", "steps", "warmup", "ms_per_step",
+    "higher_is_better", "scaling", "vs_baseline", "dtype", "data", "config",
+]
+
+
+def _check_json_line(stdout: str, n_gpus: int):
+    lines = [l for l in stdout.strip().splitlines()
+             if l.startswith("{")]
+    assert len(lines) == 1, f"expected 1 JSON line, got: {stdout!r}"
+    result = json.loads(lines[0])
+    for field in REQUIRED_FIELDS:
+        assert field in result, f"missing field {field}"
+    assert result["n_gpus"] == n_gpus
+    assert result["value"] > 0
+    assert result["higher_is_better"] is True
+    assert result["scaling"] == "weak"
+    assert result["data"] == "synthetic"
+    assert result["config"]["parallelism"] == f"dp{n_gpus}"
+    assert result["config"]["global_batch"] == \
+        result["config"]["per_gpu_batch"] * n_gpus
+    return result
+
+
+@pytest.mark.timeout(300)
+def test_bench_single_process():
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+         "--batch", "64", "--table-rows", "1000", "--dtype", "fp32"],
+        cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert out.returncode == 0, out.stderr[-2000:]
+    _check_json_line(out.stdout, 1)
+
+
+@pytest.mark.timeout(300)
+def test_bench_under_torchrun_2_ranks():
+    env = dict(os.environ)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29517", "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1", "--batch", "64",
+         "--table-rows", "1000", "--dtype", "fp32"],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=280)
+    assert out.returncode == 0, out.stderr[-2000:]
+    _check_json_line(out.stdout, 2)
