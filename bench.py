@@ -117,8 +117,10 @@ def main() -> None:
         logits = ddp(dense, ids)
         loss = loss_fn(logits.float(), labels)
         loss.backward()
+        # sparse allgather overlaps with the dense optimizer step
+        module.start_sparse_sync()
         opt.step()
-        module.apply_sparse_updates(lr)
+        module.finish_sparse_sync(lr)
         return loss
 
     log(f"rank {rank}/{world_size} device={device} "

@@ -94,14 +94,13 @@ class SparseEmbedding(nn.Module):
         return self._sink
 
     @torch.no_grad()
-    def apply_sparse_updates(self, lr: float,
-                             process_group=None) -> None:
-        """Allgather sparse grads across DP ranks and apply the fused
-        scatter+SGD update, scaled by 1/world_size."""
-        if not self._sink:
-            return
+    def start_sparse_sync(self, process_group=None) -> None:
+        """Launch the cross-rank allgather of (ids, grad rows)
+        asynchronously (overlaps with the dense optimizer step)."""
         world = (dist.get_world_size(process_group)
                  if dist.is_available() and dist.is_initialized() else 1)
+        self._pending_world = world
+        self._gathered: List[Tuple[torch.Tensor, torch.Tensor, list]] = []
         for flat_ids, grad in self._sink:
             grad2d = grad.reshape(flat_ids.numel(), self.dim)
             if world > 1:
@@ -111,18 +110,44 @@ class SparseEmbedding(nn.Module):
                 all_grads = torch.empty((n * world, self.dim),
                                         dtype=grad2d.dtype,
                                         device=grad2d.device)
-                dist.all_gather_into_tensor(all_ids, flat_ids,
-                                            group=process_group)
-                dist.all_gather_into_tensor(all_grads, grad2d,
-                                            group=process_group)
+                works = [
+                    dist.all_gather_into_tensor(all_ids, flat_ids,
+                                                group=process_group,
+                                                async_op=True),
+                    dist.all_gather_into_tensor(all_grads, grad2d,
+                                                group=process_group,
+                                                async_op=True),
+                ]
+                self._gathered.append((all_ids, all_grads, works))
             else:
-                all_ids, all_grads = flat_ids, grad2d
+                self._gathered.append((flat_ids, grad2d, []))
+        self._sink.clear()
+
+    @torch.no_grad()
+    def finish_sparse_sync(self, lr: float) -> None:
+        """Wait for the allgathers and apply the fused scatter+SGD update
+        scaled by 1/world_size."""
+        world = getattr(self, "_pending_world", 1)
+        for all_ids, all_grads, works in getattr(self, "_gathered", []):
+            for w in works:
+                w.wait()
             ops.emb_bwd_sgd(self.weight.data, all_ids, all_grads,
                             lr=lr, scale=1.0 / world)
-        self._sink.clear()
+        self._gathered = []
+
+    @torch.no_grad()
+    def apply_sparse_updates(self, lr: float,
+                             process_group=None) -> None:
+        """Allgather sparse grads across DP ranks and apply the fused
+        scatter+SGD update, scaled by 1/world_size."""
+        if not self._sink:
+            return
+        self.start_sparse_sync(process_group)
+        self.finish_sparse_sync(lr)
 
     def clear_pending(self) -> None:
         self._sink.clear()
+        self._gathered = []
 
 
 class _GatherSum(torch.autograd.Function):
@@ -171,11 +196,11 @@ class WideScalarEmbedding(nn.Module):
         return self._sink
 
     @torch.no_grad()
-    def apply_sparse_updates(self, lr: float, process_group=None) -> None:
-        if not self._sink:
-            return
+    def start_sparse_sync(self, process_group=None) -> None:
         world = (dist.get_world_size(process_group)
                  if dist.is_available() and dist.is_initialized() else 1)
+        self._pending_world = world
+        self._gathered = []
         for flat_ids, grad in self._sink:
             if world > 1:
                 n = flat_ids.numel()
@@ -184,19 +209,41 @@ class WideScalarEmbedding(nn.Module):
                                       device=flat_ids.device)
                 all_grads = torch.empty(b * world, dtype=grad.dtype,
                                         device=grad.device)
-                dist.all_gather_into_tensor(all_ids, flat_ids,
-                                            group=process_group)
-                dist.all_gather_into_tensor(all_grads, grad.reshape(-1),
-                                            group=process_group)
+                works = [
+                    dist.all_gather_into_tensor(all_ids, flat_ids,
+                                                group=process_group,
+                                                async_op=True),
+                    dist.all_gather_into_tensor(all_grads,
+                                                grad.reshape(-1),
+                                                group=process_group,
+                                                async_op=True),
+                ]
+                self._gathered.append((all_ids, all_grads, works))
             else:
-                all_ids, all_grads = flat_ids, grad.reshape(-1)
+                self._gathered.append((flat_ids, grad.reshape(-1), []))
+        self._sink.clear()
+
+    @torch.no_grad()
+    def finish_sparse_sync(self, lr: float) -> None:
+        world = getattr(self, "_pending_world", 1)
+        for all_ids, all_grads, works in getattr(self, "_gathered", []):
+            for w in works:
+                w.wait()
             ops.emb_scatter_sum(self.weight.data,
                                 all_ids.reshape(all_grads.numel(), -1),
                                 all_grads, alpha=-lr / world)
-        self._sink.clear()
+        self._gathered = []
+
+    @torch.no_grad()
+    def apply_sparse_updates(self, lr: float, process_group=None) -> None:
+        if not self._sink:
+            return
+        self.start_sparse_sync(process_group)
+        self.finish_sparse_sync(lr)
 
     def clear_pending(self) -> None:
         self._sink.clear()
+        self._gathered = []
 
 
 class FusedLinearReLU(nn.Module):
@@ -257,9 +304,20 @@ class WideAndDeep(nn.Module):
                     + self.wide_dense(dense))
         return (deep_out + wide_out).squeeze(1)
 
+    def start_sparse_sync(self, process_group=None) -> None:
+        """Kick off both embeddings' allgathers (call right after
+        ``loss.backward()`` so the communication overlaps with the dense
+        optimizer step)."""
+        self.deep_embedding.start_sparse_sync(process_group)
+        self.wide_embedding.start_sparse_sync(process_group)
+
+    def finish_sparse_sync(self, lr: float) -> None:
+        self.deep_embedding.finish_sparse_sync(lr)
+        self.wide_embedding.finish_sparse_sync(lr)
+
     def apply_sparse_updates(self, lr: float, process_group=None) -> None:
-        self.deep_embedding.apply_sparse_updates(lr, process_group)
-        self.wide_embedding.apply_sparse_updates(lr, process_group)
+        self.start_sparse_sync(process_group)
+        self.finish_sparse_sync(lr)
 
     def clear_pending(self) -> None:
         self.deep_embedding.clear_pending()
