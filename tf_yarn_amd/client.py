@@ -120,7 +120,7 @@ def _allocate_gpus(tasks: List[ContainerTask],
         spec = task_specs.get(t.type)
         key = f"{t.type}:{t.id}"
         if (spec is not None and spec.label == NodeLabel.GPU
-                and t.type in ("chief", "worker")):
+                and t.type in ("chief", "worker", "ps")):
             gpus = [(next_gpu + i) % constants.NODE_GPU_COUNT
                     for i in range(t.nb_proc)]
             next_gpu = (next_gpu + t.nb_proc) % constants.NODE_GPU_COUNT

@@ -21,6 +21,7 @@ MAX_VCORES = 256
 
 ALL_TASK_TYPES = {"chief", "worker", "ps", "evaluator", "tensorboard"}
 TRAINING_TASK_TYPES = {"chief", "worker"}
+GPU_ELIGIBLE_TASK_TYPES = {"chief", "worker", "ps"}
 
 
 class NodeLabel(Enum):
@@ -117,7 +118,7 @@ def _check_general_topology(task_specs: TaskSpecs) -> None:
     n_gpu_procs = sum(
         spec.instances * spec.nb_proc_per_worker
         for t, spec in task_specs.items()
-        if spec.label == NodeLabel.GPU and t in TRAINING_TASK_TYPES)
+        if spec.label == NodeLabel.GPU and t in GPU_ELIGIBLE_TASK_TYPES)
     if n_gpu_procs > constants.NODE_GPU_COUNT:
         raise ValueError(
             f"{n_gpu_procs} GPU training processes requested but the node "
