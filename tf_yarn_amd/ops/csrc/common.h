@@ -39,6 +39,14 @@ __device__ __forceinline__ unsigned short f32_to_bf16(float f) {
   return static_cast<unsigned short>(x >> 16);
 }
 
+// Native fp32 global atomic add: plain atomicAdd lowers to a CAS loop on
+// ROCm without -munsafe-fp-atomics; unsafeAtomicAdd emits
+// global_atomic_add_f32 directly (correct on HBM; the "unsafe" caveat is
+// about fine-grained host memory, which these kernels never touch).
+__device__ __forceinline__ void f32_atomic_add(float* p, float v) {
+  unsafeAtomicAdd(p, v);
+}
+
 // packed vectors: 16 B per lane
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(8))) unsigned short bf16x8;

@@ -101,7 +101,7 @@ __global__ void bias_relu_bwd_db_kernel(
   for (int64_t q = threadIdx.x; q < quads; q += blockDim.x, ++k) {
 #pragma unroll
     for (int j = 0; j < 4; ++j)
-      atomicAdd(&dbias[q * 4 + j], acc[k][j]);
+      f32_atomic_add(&dbias[q * 4 + j], acc[k][j]);
   }
 }
 

@@ -64,7 +64,7 @@ __global__ void emb_bwd_sgd_kernel(float* __restrict__ table,
     QuadIo<GIo>::load4(g, row * dvec + c4, gv);
 #pragma unroll
     for (int j = 0; j < 4; ++j)
-      atomicAdd(dst + j, neg_lr_scale * gv[j]);
+      f32_atomic_add(dst + j, neg_lr_scale * gv[j]);
   }
 }
 
@@ -86,7 +86,7 @@ __global__ void emb_bwd_dense_kernel(
     QuadIo<GIo>::load4(g, row * dvec + c4, gv);
 #pragma unroll
     for (int j = 0; j < 4; ++j)
-      atomicAdd(dst + j, scale * gv[j]);
+      f32_atomic_add(dst + j, scale * gv[j]);
   }
 }
 
@@ -117,7 +117,7 @@ __global__ void emb_scatter_scalar_kernel(
        t < total; t += stride) {
     const int64_t row = t / dim;
     const int64_t c = t - row * dim;
-    atomicAdd(table + ids[row] * dim + c, alpha * GIo::load(g, t));
+    f32_atomic_add(table + ids[row] * dim + c, alpha * GIo::load(g, t));
   }
 }
 
@@ -150,7 +150,7 @@ __global__ void emb_scatter_sum_kernel(float* __restrict__ table,
   for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
        t < total; t += stride) {
     const int64_t b = t / F;
-    atomicAdd(table + ids[t], alpha * GIo::load(g, b));
+    f32_atomic_add(table + ids[t], alpha * GIo::load(g, b));
   }
 }
 
