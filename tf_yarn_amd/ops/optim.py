@@ -26,11 +26,26 @@ def _split_param(p: torch.Tensor, state: dict):
     return p.data, None
 
 
+def _flat(t: torch.Tensor) -> torch.Tensor:
+    """Flat view in STORAGE order (handles channels_last params, whose
+    logical view(-1) is not a view of storage)."""
+    if t.is_contiguous():
+        return t.view(-1)
+    return torch.as_strided(t, (t.numel(),), (1,),
+                            storage_offset=t.storage_offset())
+
+
 def _grad_of(p: torch.Tensor) -> Optional[torch.Tensor]:
+    """Gradient with the SAME memory layout as the param (elementwise
+    kernels walk storage order, so layouts must match)."""
     g = p.grad
     if g is None:
         return None
-    return g.contiguous()
+    if g.stride() == p.data.stride():
+        return g
+    aligned = torch.empty_like(p.data, dtype=g.dtype)
+    aligned.copy_(g)
+    return aligned
 
 
 class FusedSGD(Optimizer):
@@ -67,12 +82,12 @@ class FusedSGD(Optimizer):
                 key = (g.dtype, mom_buf is not None, bf16 is not None,
                        first)
                 b = buckets.setdefault(key, ([], [], [], []))
-                b[0].append(master.view(-1))
-                b[1].append(g.view(-1))
+                b[0].append(_flat(master))
+                b[1].append(_flat(g))
                 if mom_buf is not None:
-                    b[2].append(mom_buf.view(-1))
+                    b[2].append(_flat(mom_buf))
                 if bf16 is not None:
-                    b[3].append(bf16.view(-1))
+                    b[3].append(_flat(bf16))
                 if bf16 is None and p.dtype != torch.float32:
                     p.data.copy_(master.to(p.dtype))
             for (gdtype, has_mom, has_bf16, first), \
@@ -117,10 +132,10 @@ class FusedAdam(Optimizer):
                     state["exp_avg_sq"] = torch.zeros_like(master)
                 state["step"] = state.get("step", 0) + 1
                 ops.fused_adam(
-                    master.view(-1), g.view(-1),
-                    state["exp_avg"].view(-1),
-                    state["exp_avg_sq"].view(-1),
-                    bf16.view(-1) if bf16 is not None else None,
+                    _flat(master), _flat(g),
+                    _flat(state["exp_avg"]),
+                    _flat(state["exp_avg_sq"]),
+                    _flat(bf16) if bf16 is not None else None,
                     lr=group["lr"], beta1=beta1, beta2=beta2,
                     eps=group["eps"], weight_decay=group["weight_decay"],
                     adamw=group["adamw"], step=state["step"],
@@ -151,7 +166,7 @@ class FusedAdagrad(Optimizer):
                 if "sum" not in state:
                     state["sum"] = torch.zeros_like(master)
                 ops.fused_adagrad(
-                    master.view(-1), g.view(-1), state["sum"].view(-1),
+                    _flat(master), _flat(g), _flat(state["sum"]),
                     lr=group["lr"], eps=group["eps"],
                     weight_decay=group["weight_decay"],
                     grad_scale=group["grad_scale"])
@@ -187,9 +202,9 @@ class FusedAdadelta(Optimizer):
                     state["square_avg"] = torch.zeros_like(master)
                     state["acc_delta"] = torch.zeros_like(master)
                 ops.fused_adadelta(
-                    master.view(-1), g.view(-1),
-                    state["square_avg"].view(-1),
-                    state["acc_delta"].view(-1),
+                    _flat(master), _flat(g),
+                    _flat(state["square_avg"]),
+                    _flat(state["acc_delta"]),
                     lr=group["lr"], rho=group["rho"], eps=group["eps"],
                     weight_decay=group["weight_decay"],
                     grad_scale=group["grad_scale"])
