@@ -258,3 +258,67 @@ def test_emb_gather_scatter_sum_gpu():
     expanded = grad.reshape(-1, 1).expand(-1, 26).reshape(-1)
     ref2.reshape(-1).index_add_(0, ids.reshape(-1), expanded, alpha=-0.1)
     assert torch.allclose(table2, ref2, atol=1e-3)
+
+
+@requires_gpu
+def test_run_on_yarn_gpu_single():
+    """End-to-end spawner run on the GPU box: chief task pinned to GPU 0,
+    RCCL process group (world_size 1), HIP kernels in the training step."""
+    import os
+    import sys
+    import tempfile
+
+    import cloudpickle
+
+    cloudpickle.register_pickle_by_value(sys.modules[__name__])
+    from tf_yarn_amd import NodeLabel, TaskSpec
+    from tf_yarn_amd.pytorch import run_on_yarn
+
+    model_dir = tempfile.mkdtemp(prefix="miyarn_gpu_e2e_")
+
+    def experiment_fn():
+        import torch
+        from torch import nn
+
+        from tf_yarn_amd.models.synthetic import SyntheticCriteoDataset
+        from tf_yarn_amd.models.wide_deep import WideAndDeep
+        from tf_yarn_amd.ops.optim import FusedSGD
+        from tf_yarn_amd.pytorch import DataLoaderArgs, PytorchExperiment
+
+        tables = [1000] * 26
+
+        def main_fn(model, loader, device, rank, tb_writer):
+            module = model.module if hasattr(model, "module") else model
+            assert device.startswith("cuda"), f"expected GPU, got {device}"
+            opt = FusedSGD([p for p in module.parameters()
+                            if not getattr(p, "_miyarn_sparse", False)],
+                           lr=0.05)
+            for dense, ids, labels in loader:
+                dense = dense[0].to(device)
+                ids = ids[0].to(device)
+                labels = labels.reshape(-1).to(device)
+                opt.zero_grad(set_to_none=False)
+                loss = nn.functional.binary_cross_entropy_with_logits(
+                    model(dense, ids).float(), labels)
+                loss.backward()
+                opt.step()
+                module.apply_sparse_updates(0.05)
+            assert torch.isfinite(loss)
+
+        torch.manual_seed(0)
+        model = WideAndDeep(table_sizes=tables, embedding_dim=16,
+                            hidden=(64, 32),
+                            compute_dtype=torch.bfloat16)
+        return PytorchExperiment(
+            model=model, main_fn=main_fn,
+            train_dataset=SyntheticCriteoDataset(8 * 64, tables,
+                                                 batch_size=64),
+            dataloader_args=DataLoaderArgs(batch_size=1,
+                                           pin_memory=False))
+
+    metrics = run_on_yarn(
+        experiment_fn,
+        {"chief": TaskSpec(memory=2048, vcores=1, label=NodeLabel.GPU)},
+        base_dir=model_dir)
+    assert metrics is not None
+    assert metrics.total_training_duration is not None
