@@ -111,9 +111,13 @@ def main() -> None:
             args.batch, table_sizes, device=device, seed=1000 * rank + i)
         batches.append((dense, ids, labels))
 
+    # world_size==1 has no bucket views: set_to_none avoids the fill+
+    # accumulate kernels entirely (grads are assigned fresh each step)
+    set_to_none = world_size == 1
+
     def step(i: int) -> float:
         dense, ids, labels = batches[i % N_DATA_BATCHES]
-        opt.zero_grad(set_to_none=False)
+        opt.zero_grad(set_to_none=set_to_none)
         logits = ddp(dense, ids)
         loss = loss_fn(logits.float(), labels)
         loss.backward()

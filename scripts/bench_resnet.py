@@ -35,7 +35,9 @@ def main() -> None:
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--batch", type=int, default=PER_GPU_BATCH)
-    ap.add_argument("--channels-last", action="store_true")
+    # channels_last measured 6.6k vs 4.2k img/s NCHW on MI355X: default on
+    ap.add_argument("--channels-last", action="store_true", default=True)
+    ap.add_argument("--nchw", dest="channels_last", action="store_false")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))

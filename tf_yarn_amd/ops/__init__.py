@@ -209,6 +209,22 @@ def convert_scaled(src: torch.Tensor, dst: torch.Tensor,
     dst.copy_((src.float() * scale).to(dst.dtype))
 
 
+def emb_fwd_into(table: torch.Tensor, ids: torch.Tensor,
+                 out: torch.Tensor, col_offset: int) -> None:
+    """Gather row (b,f) into out[b, col_offset + f*dim : +dim]."""
+    if _on_gpu(table, ids, out):
+        _require_ext()
+        _C.emb_fwd_into(table, ids.reshape(-1).contiguous(), out,
+                        col_offset)
+        return
+    batch = out.shape[0]
+    f = ids.numel() // batch
+    dim = table.shape[1]
+    gathered = table.index_select(0, ids.reshape(-1)).reshape(
+        batch, f * dim)
+    out[:, col_offset:col_offset + f * dim] = gathered.to(out.dtype)
+
+
 def emb_gather_sum(table: torch.Tensor, ids: torch.Tensor,
                    out_bf16: bool = False) -> torch.Tensor:
     """out[b] = sum_f table[ids[b, f], 0] for a scalar (dim-1) table."""
@@ -249,11 +265,14 @@ class BiasReLU(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         (y,) = ctx.saved_tensors
-        # Measured (micro_elementwise.py @65536x1024): separate dx kernel
-        # (82 us) + torch reduce (44 us) beats the fused dx+dbias kernel
-        # (268 us, atomic-contention-bound: one atomic per column per
-        # block).  Keep the separate path; _C.bias_relu_bwd_db stays
-        # available for low-row-count shapes.
+        if dy.is_cuda and HAVE_EXT and y.size(-1) % 4 == 0 \
+                and y.size(-1) <= 8192:
+            # v3 fused kernel: atomic-free per-block column partials +
+            # one tiny reduce (earlier atomic/LDS variants measured
+            # slower than the separate path; see micro_elementwise.py).
+            dx, dbias32 = _C.bias_relu_bwd_db(dy.contiguous(),
+                                              y.contiguous())
+            return dx, dbias32.to(dy.dtype)
         dx = bias_relu_bwd(dy.contiguous(), y)
         dims = tuple(range(dx.dim() - 1))
         dbias = dx.sum(dim=dims)

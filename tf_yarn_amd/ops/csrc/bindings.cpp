@@ -39,6 +39,8 @@ void emb_bwd_dense(torch::Tensor grad_table, torch::Tensor ids,
 
 torch::Tensor emb_gather_sum(torch::Tensor table, torch::Tensor ids,
                              int64_t batch, bool out_bf16);
+void emb_fwd_into(torch::Tensor table, torch::Tensor ids,
+                  torch::Tensor out, int64_t col_offset);
 void emb_scatter_sum(torch::Tensor table, torch::Tensor ids,
                      torch::Tensor grad, double alpha);
 
@@ -66,6 +68,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_relu_bwd", &bias_relu_bwd, "Fused ReLU backward");
   m.def("bias_relu_bwd_db", &bias_relu_bwd_db,
         "Fused ReLU backward + dbias reduction (returns [dx, dbias_fp32])");
+  m.def("emb_fwd_into", &emb_fwd_into,
+        "Embedding gather into a slice of a larger 2D buffer");
   m.def("emb_gather_sum", &emb_gather_sum,
         "Wide-part gather-sum: out[b] = sum_f table[ids[b,f]]");
   m.def("emb_scatter_sum", &emb_scatter_sum,

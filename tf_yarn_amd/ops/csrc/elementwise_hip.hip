@@ -31,6 +31,103 @@ __global__ void bias_relu_fwd_kernel(
   }
 }
 
+// Vectorized variant (cols % 4 == 0): one quad per iteration, 8-16 B/lane.
+template <typename Io>
+__global__ void bias_relu_fwd_vec_kernel(
+    const typename Io::scalar_t* __restrict__ x,
+    const typename Io::scalar_t* __restrict__ bias,
+    typename Io::scalar_t* __restrict__ y,
+    int64_t total_quads, int64_t qcols) {
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       q < total_quads; q += stride) {
+    const int64_t cq = q % qcols;
+    float v[4], b[4];
+    QuadIo<Io>::load4(x, q, v);
+    QuadIo<Io>::load4(bias, cq, b);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      v[j] += b[j];
+      v[j] = v[j] > 0.f ? v[j] : 0.f;
+    }
+    QuadIo<Io>::store4(y, q, v);
+  }
+}
+
+// Vectorized ReLU backward: dx = dy * (y > 0), quads.
+template <typename Io>
+__global__ void bias_relu_bwd_vec_kernel(
+    const typename Io::scalar_t* __restrict__ dy,
+    const typename Io::scalar_t* __restrict__ y,
+    typename Io::scalar_t* __restrict__ dx,
+    int64_t total_quads) {
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       q < total_quads; q += stride) {
+    float g[4], yy[4];
+    QuadIo<Io>::load4(dy, q, g);
+    QuadIo<Io>::load4(y, q, yy);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) g[j] = yy[j] > 0.f ? g[j] : 0.f;
+    QuadIo<Io>::store4(dx, q, g);
+  }
+}
+
+// Fused dx + per-block dbias PARTIALS (atomic-free): block b owns rows
+// [b*DB_ROWS :: grid*DB_ROWS]; thread partials in registers; at the end
+// each block writes its own row of dbias_part[grid][cols], which the host
+// reduces with one tiny torch sum (2048 x cols).
+#define DB_MAX_K 8  // max column-quads per thread (cols <= 4*block*K)
+#define DB_ROWS 4   // rows in flight per thread (ILP for HBM latency)
+
+template <typename Io>
+__global__ void bias_relu_bwd_dbpart_kernel(
+    const typename Io::scalar_t* __restrict__ dy,
+    const typename Io::scalar_t* __restrict__ y,
+    typename Io::scalar_t* __restrict__ dx,
+    float* __restrict__ dbias_part,
+    int64_t rows, int64_t cols) {
+  const int64_t quads = cols >> 2;
+  float acc[DB_MAX_K][4];
+#pragma unroll
+  for (int k = 0; k < DB_MAX_K; ++k)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[k][j] = 0.f;
+  const int64_t row_stride = (int64_t)gridDim.x * DB_ROWS;
+  for (int64_t r0 = (int64_t)blockIdx.x * DB_ROWS; r0 < rows;
+       r0 += row_stride) {
+    const int nr = min((int64_t)DB_ROWS, rows - r0);
+    int k = 0;
+    for (int64_t q = threadIdx.x; q < quads; q += blockDim.x, ++k) {
+      float v[DB_ROWS][4], yy[DB_ROWS][4];
+#pragma unroll
+      for (int rr = 0; rr < DB_ROWS; ++rr)
+        if (rr < nr) QuadIo<Io>::load4(dy, (r0 + rr) * quads + q, v[rr]);
+#pragma unroll
+      for (int rr = 0; rr < DB_ROWS; ++rr)
+        if (rr < nr) QuadIo<Io>::load4(y, (r0 + rr) * quads + q, yy[rr]);
+#pragma unroll
+      for (int rr = 0; rr < DB_ROWS; ++rr)
+        if (rr < nr) {
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            v[rr][j] = yy[rr][j] > 0.f ? v[rr][j] : 0.f;
+            acc[k][j] += v[rr][j];
+          }
+          QuadIo<Io>::store4(dx, (r0 + rr) * quads + q, v[rr]);
+        }
+    }
+  }
+  float* part = dbias_part + (int64_t)blockIdx.x * cols;
+  int k = 0;
+  for (int64_t q = threadIdx.x; q < quads; q += blockDim.x, ++k) {
+    f32x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o[j] = acc[k][j];
+    reinterpret_cast<f32x4*>(part)[q] = o;
+  }
+}
+
 // dx = dy * (y > 0)
 template <typename Io>
 __global__ void bias_relu_bwd_kernel(
@@ -44,65 +141,6 @@ __global__ void bias_relu_bwd_kernel(
     float g = Io::load(dy, i);
     float yy = Io::load(y, i);
     Io::store(dx, i, yy > 0.f ? g : 0.f);
-  }
-}
-
-// dx = dy * (y > 0) AND dbias[c] += sum_rows dx  (fused: the separate
-// torch bf16 reduce was 23 us x 3 layers per step in the profile).
-// Row-wise: each thread owns 4 consecutive columns (8 B/lane bf16 loads,
-// coalesced), accumulates its column partials in registers over its
-// block's rows, then does ONE atomicAdd per owned column.  Column count
-// per thread = cols / (4 * blockDim) rounded up, capped at DB_MAX_K.
-#define DB_MAX_K 8
-
-#define DB_ROWS 4  // rows in flight per thread (ILP for the HBM latency)
-
-template <typename Io>
-__global__ void bias_relu_bwd_db_kernel(
-    const typename Io::scalar_t* __restrict__ dy,
-    const typename Io::scalar_t* __restrict__ y,
-    typename Io::scalar_t* __restrict__ dx,
-    float* __restrict__ dbias,
-    int64_t rows, int64_t cols) {
-  const int64_t quads = cols >> 2;  // cols % 4 == 0 guaranteed by host
-  float acc[DB_MAX_K][4];
-#pragma unroll
-  for (int k = 0; k < DB_MAX_K; ++k)
-#pragma unroll
-    for (int j = 0; j < 4; ++j) acc[k][j] = 0.f;
-  const int64_t row_stride = (int64_t)gridDim.x * DB_ROWS;
-  for (int64_t r0 = (int64_t)blockIdx.x * DB_ROWS; r0 < rows;
-       r0 += row_stride) {
-    const int nr = min((int64_t)DB_ROWS, rows - r0);
-    int k = 0;
-    for (int64_t q = threadIdx.x; q < quads; q += blockDim.x, ++k) {
-      float v[DB_ROWS][4];
-#pragma unroll
-      for (int rr = 0; rr < DB_ROWS; ++rr)
-        if (rr < nr)
-          QuadIo<Io>::load4(dy, (r0 + rr) * quads + q, v[rr]);
-      float yy[DB_ROWS][4];
-#pragma unroll
-      for (int rr = 0; rr < DB_ROWS; ++rr)
-        if (rr < nr)
-          QuadIo<Io>::load4(y, (r0 + rr) * quads + q, yy[rr]);
-#pragma unroll
-      for (int rr = 0; rr < DB_ROWS; ++rr)
-        if (rr < nr) {
-#pragma unroll
-          for (int j = 0; j < 4; ++j) {
-            v[rr][j] = yy[rr][j] > 0.f ? v[rr][j] : 0.f;
-            acc[k][j] += v[rr][j];
-          }
-          QuadIo<Io>::store4(dx, (r0 + rr) * quads + q, v[rr]);
-        }
-    }
-  }
-  int k = 0;
-  for (int64_t q = threadIdx.x; q < quads; q += blockDim.x, ++k) {
-#pragma unroll
-    for (int j = 0; j < 4; ++j)
-      f32_atomic_add(&dbias[q * 4 + j], acc[k][j]);
   }
 }
 
@@ -151,6 +189,26 @@ torch::Tensor bias_relu_fwd(torch::Tensor x, torch::Tensor bias) {
   TORCH_CHECK(bias.numel() == cols, "bias size mismatch");
   auto y = torch::empty_like(x);
   auto stream = c10::hip::getCurrentHIPStream().stream();
+  const bool vec = (cols % 4 == 0);
+  if (vec) {
+    const int64_t tq = rows * (cols / 4);
+    int grid = miyarn_grid(tq);
+    if (x.scalar_type() == torch::kFloat32) {
+      hipLaunchKernelGGL(bias_relu_fwd_vec_kernel<F32Io>, dim3(grid),
+                         dim3(MIYARN_BLOCK), 0, stream,
+                         x.data_ptr<float>(), bias.data_ptr<float>(),
+                         y.data_ptr<float>(), tq, cols / 4);
+    } else {
+      TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "fp32/bf16 only");
+      hipLaunchKernelGGL(bias_relu_fwd_vec_kernel<Bf16Io>, dim3(grid),
+                         dim3(MIYARN_BLOCK), 0, stream,
+                         reinterpret_cast<unsigned short*>(x.data_ptr()),
+                         reinterpret_cast<unsigned short*>(bias.data_ptr()),
+                         reinterpret_cast<unsigned short*>(y.data_ptr()),
+                         tq, cols / 4);
+    }
+    return y;
+  }
   int grid = miyarn_grid(rows * cols);
   if (x.scalar_type() == torch::kFloat32) {
     hipLaunchKernelGGL(bias_relu_fwd_kernel<F32Io>, dim3(grid),
@@ -177,6 +235,25 @@ torch::Tensor bias_relu_bwd(torch::Tensor dy, torch::Tensor y) {
   auto dx = torch::empty_like(dy);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   int64_t total = dy.numel();
+  if (total % 4 == 0) {
+    const int64_t tq = total / 4;
+    int grid = miyarn_grid(tq);
+    if (dy.scalar_type() == torch::kFloat32) {
+      hipLaunchKernelGGL(bias_relu_bwd_vec_kernel<F32Io>, dim3(grid),
+                         dim3(MIYARN_BLOCK), 0, stream,
+                         dy.data_ptr<float>(), y.data_ptr<float>(),
+                         dx.data_ptr<float>(), tq);
+    } else {
+      TORCH_CHECK(dy.scalar_type() == torch::kBFloat16, "fp32/bf16 only");
+      hipLaunchKernelGGL(bias_relu_bwd_vec_kernel<Bf16Io>, dim3(grid),
+                         dim3(MIYARN_BLOCK), 0, stream,
+                         reinterpret_cast<unsigned short*>(dy.data_ptr()),
+                         reinterpret_cast<unsigned short*>(y.data_ptr()),
+                         reinterpret_cast<unsigned short*>(dx.data_ptr()),
+                         tq);
+    }
+    return dx;
+  }
   int grid = miyarn_grid(total);
   if (dy.scalar_type() == torch::kFloat32) {
     hipLaunchKernelGGL(bias_relu_bwd_kernel<F32Io>, dim3(grid),
@@ -206,26 +283,32 @@ std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
               "bias_relu_bwd_db needs cols % 4 == 0 and cols <= ",
               4 * MIYARN_BLOCK * DB_MAX_K);
   auto dx = torch::empty_like(dy);
-  auto dbias = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
   auto stream = c10::hip::getCurrentHIPStream().stream();
+  // Block size matched to the quad count so no thread idles at small cols.
+  const int64_t quads = cols / 4;
+  int block = static_cast<int>(
+      std::min<int64_t>(MIYARN_BLOCK, ((quads + 63) / 64) * 64));
   int grid = static_cast<int>(std::min<int64_t>(
       (rows + DB_ROWS - 1) / DB_ROWS, MIYARN_MAX_BLOCKS));
-  size_t lds = 0;
+  // Atomic-free column partials: one row per block, reduced below.
+  auto part = torch::empty({grid, cols},
+                           dy.options().dtype(torch::kFloat32));
   if (dy.scalar_type() == torch::kFloat32) {
-    hipLaunchKernelGGL(bias_relu_bwd_db_kernel<F32Io>, dim3(grid),
-                       dim3(MIYARN_BLOCK), lds, stream,
+    hipLaunchKernelGGL(bias_relu_bwd_dbpart_kernel<F32Io>, dim3(grid),
+                       dim3(block), 0, stream,
                        dy.data_ptr<float>(), y.data_ptr<float>(),
-                       dx.data_ptr<float>(), dbias.data_ptr<float>(),
+                       dx.data_ptr<float>(), part.data_ptr<float>(),
                        rows, cols);
   } else {
     TORCH_CHECK(dy.scalar_type() == torch::kBFloat16, "fp32/bf16 only");
-    hipLaunchKernelGGL(bias_relu_bwd_db_kernel<Bf16Io>, dim3(grid),
-                       dim3(MIYARN_BLOCK), lds, stream,
+    hipLaunchKernelGGL(bias_relu_bwd_dbpart_kernel<Bf16Io>, dim3(grid),
+                       dim3(block), 0, stream,
                        reinterpret_cast<unsigned short*>(dy.data_ptr()),
                        reinterpret_cast<unsigned short*>(y.data_ptr()),
                        reinterpret_cast<unsigned short*>(dx.data_ptr()),
-                       dbias.data_ptr<float>(), rows, cols);
+                       part.data_ptr<float>(), rows, cols);
   }
+  auto dbias = part.sum(0);
   return {dx, dbias};
 }
 

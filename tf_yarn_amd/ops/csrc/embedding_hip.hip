@@ -154,6 +154,34 @@ __global__ void emb_scatter_sum_kernel(float* __restrict__ table,
   }
 }
 
+// Gather straight into a slice of the MLP input buffer: gather row
+// (b, f) = ids[b*F + f] lands at out[b, col_off + f*dim : ...], killing
+// the separate concat kernel (out stride/offset in quads; host guarantees
+// dim % 4 == 0 and col_off % 4 == 0).
+template <typename OIo>
+__global__ void emb_fwd_into_kernel(const float* __restrict__ table,
+                                    const int64_t* __restrict__ ids,
+                                    typename OIo::scalar_t* __restrict__ out,
+                                    int64_t n_rows, int64_t dim, int F,
+                                    int64_t out_stride_q, int64_t off_q) {
+  const int64_t dvec = dim >> 2;
+  const int64_t total = n_rows * dvec;
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       t < total; t += stride) {
+    const int64_t row = t / dvec;
+    const int64_t c4 = t - row * dvec;
+    const int64_t b = row / F;
+    const int64_t f = row - b * F;
+    f32x4 v = reinterpret_cast<const f32x4*>(table + ids[row] * dim)[c4];
+    float vv[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) vv[j] = v[j];
+    QuadIo<OIo>::store4(out, b * out_stride_q + off_q + f * dvec + c4,
+                        vv);
+  }
+}
+
 void check_emb(const torch::Tensor& table, const torch::Tensor& ids,
                int64_t dim) {
   TORCH_CHECK(table.is_cuda() && table.is_contiguous() &&
@@ -338,5 +366,38 @@ void emb_scatter_sum(torch::Tensor table, torch::Tensor ids,
                        table.data_ptr<float>(), ids.data_ptr<int64_t>(),
                        reinterpret_cast<unsigned short*>(grad.data_ptr()),
                        batch, F, (float)alpha);
+  }
+}
+
+void emb_fwd_into(torch::Tensor table, torch::Tensor ids,
+                  torch::Tensor out, int64_t col_offset) {
+  const int64_t dim = table.size(1);
+  const int64_t n = ids.numel();
+  check_emb(table, ids, dim);
+  TORCH_CHECK(out.is_cuda() && out.is_contiguous() && out.dim() == 2,
+              "out must be a contiguous 2D GPU tensor");
+  TORCH_CHECK(dim % 4 == 0 && col_offset % 4 == 0 &&
+              out.size(1) % 4 == 0,
+              "emb_fwd_into needs dim, col_offset and out stride % 4 == 0");
+  const int64_t batch = out.size(0);
+  TORCH_CHECK(n % batch == 0, "ids not divisible by out rows");
+  const int F = static_cast<int>(n / batch);
+  TORCH_CHECK(col_offset + (int64_t)F * dim <= out.size(1),
+              "slice out of range");
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int grid = miyarn_grid(n * (dim / 4));
+  if (out.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL(emb_fwd_into_kernel<Bf16Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream,
+                       table.data_ptr<float>(), ids.data_ptr<int64_t>(),
+                       reinterpret_cast<unsigned short*>(out.data_ptr()),
+                       n, dim, F, out.size(1) / 4, col_offset / 4);
+  } else {
+    TORCH_CHECK(out.scalar_type() == torch::kFloat32, "fp32/bf16 only");
+    hipLaunchKernelGGL(emb_fwd_into_kernel<F32Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream,
+                       table.data_ptr<float>(), ids.data_ptr<int64_t>(),
+                       out.data_ptr<float>(), n, dim, F,
+                       out.size(1) / 4, col_offset / 4);
   }
 }
