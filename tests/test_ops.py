@@ -100,3 +100,28 @@ def test_convert_scaled_cpu():
     dst = torch.empty(13, dtype=torch.bfloat16)
     ops.convert_scaled(src, dst, 2.0)
     assert torch.allclose(dst.float(), (src * 2).to(torch.bfloat16).float())
+
+
+def test_col_reduce_dot_cpu():
+    x = torch.randn(50, 8)
+    dy = torch.randn(50)
+    out = ops.col_reduce_dot(x, dy)
+    ref = (x * dy.unsqueeze(1)).sum(dim=0)
+    assert torch.allclose(out, ref, atol=1e-5)
+
+
+def test_scalar_head_fn_matches_autograd():
+    torch.manual_seed(4)
+    x = torch.randn(32, 8, requires_grad=True)
+    w = torch.randn(8, requires_grad=True)
+    b = torch.zeros(1, requires_grad=True)
+    y = ops.ScalarHeadFn.apply(x, w, b)
+    y.pow(2).sum().backward()
+    x2 = x.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    (x2 @ w2 + b2).pow(2).sum().backward()
+    assert torch.allclose(y, x2 @ w2 + b2, atol=1e-6)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-5)
+    assert torch.allclose(w.grad, w2.grad, atol=1e-4)
+    assert torch.allclose(b.grad, b2.grad, atol=1e-5)

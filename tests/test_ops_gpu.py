@@ -322,3 +322,39 @@ def test_run_on_yarn_gpu_single():
         base_dir=model_dir)
     assert metrics is not None
     assert metrics.total_training_duration is not None
+
+
+@requires_gpu
+def test_col_reduce_dot_gpu():
+    torch.manual_seed(12)
+    for dtype in (torch.float32, torch.bfloat16):
+        x = torch.randn(10000, 256).cuda().to(dtype)
+        dy = torch.randn(10000).cuda().to(dtype)
+        out = ops.col_reduce_dot(x, dy)
+        ref = (x.float() * dy.float().unsqueeze(1)).sum(dim=0)
+        assert torch.allclose(out, ref, atol=0.5, rtol=1e-2), dtype
+
+
+@requires_gpu
+def test_emb_fwd_into_gpu():
+    torch.manual_seed(13)
+    table = torch.randn(300, 16).cuda()
+    ids = torch.randint(0, 300, (64, 26)).cuda()
+    out = torch.full((64, 16 + 26 * 16), -9.0,
+                     dtype=torch.bfloat16).cuda()
+    ops.emb_fwd_into(table, ids, out, 16)
+    ref = table.index_select(0, ids.reshape(-1)).reshape(
+        64, 26 * 16).to(torch.bfloat16)
+    assert torch.equal(out[:, 16:], ref)
+    assert (out[:, :16] == -9.0).all()  # untouched prefix
+
+
+@requires_gpu
+def test_bias_relu_fwd_vec_gpu():
+    torch.manual_seed(14)
+    for dtype in (torch.float32, torch.bfloat16):
+        x = torch.randn(1000, 512).cuda().to(dtype)
+        b = torch.randn(512).cuda().to(dtype)
+        y = ops.bias_relu_fwd(x, b)
+        ref = torch.relu(x.float() + b.float()).to(dtype)
+        assert torch.allclose(y.float(), ref.float(), atol=1e-2)

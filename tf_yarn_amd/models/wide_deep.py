@@ -294,6 +294,8 @@ class ScalarHead(nn.Module):
             self.bias.data = self.bias.data.to(dtype)
 
     def forward(self, x):
+        if x.size(-1) % 4 == 0:
+            return ops.ScalarHeadFn.apply(x, self.weight, self.bias)
         return x @ self.weight + self.bias
 
 

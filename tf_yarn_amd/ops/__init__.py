@@ -252,6 +252,34 @@ def emb_scatter_sum(table: torch.Tensor, ids: torch.Tensor,
                                  alpha=alpha)
 
 
+def col_reduce_dot(x: torch.Tensor, dy: torch.Tensor) -> torch.Tensor:
+    """dw[m] = sum_b dy[b] * x[b, m] (fp32 result)."""
+    if _on_gpu(x, dy):
+        _require_ext()
+        return _C.col_reduce_dot(x.contiguous(), dy.contiguous())
+    return (x.float() * dy.float().unsqueeze(1)).sum(dim=0)
+
+
+class ScalarHeadFn(torch.autograd.Function):
+    """y[b] = x[b].w + bias for a single-logit head; the wgrad runs as a
+    streaming column reduction instead of hipBLASLt's M=1 GEMM
+    (187 us -> ~15 us measured at 65536x256)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        ctx.save_for_backward(x, weight)
+        return x @ weight + bias
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        dy = dy.contiguous()
+        dw = col_reduce_dot(x, dy).to(weight.dtype)
+        dx = dy.unsqueeze(1) * weight.unsqueeze(0)
+        db = dy.sum().reshape(1).to(weight.dtype)
+        return dx, dw, db
+
+
 class BiasReLU(torch.autograd.Function):
     """Autograd wrapper for the fused bias+ReLU epilogue (backward fuses
     the dbias reduction into the dx kernel on GPU)."""

@@ -335,3 +335,88 @@ void convert_scaled(torch::Tensor src, torch::Tensor dst, double scale) {
     TORCH_CHECK(false, "convert_scaled supports bf16<->fp32 only");
   }
 }
+
+namespace {
+
+// dw[m] = sum_b dy[b] * x[b, m]  — the wgrad of a single-logit head as a
+// memory-bound column reduction (hipBLASLt runs this M=1-ish GEMM at
+// ~190 us for 33 MB; this kernel is a straight stream).  Atomic-free:
+// per-block partials + host-side sum, like bias_relu_bwd_dbpart.
+template <typename Io>
+__global__ void col_reduce_dot_kernel(
+    const typename Io::scalar_t* __restrict__ x,
+    const typename Io::scalar_t* __restrict__ dy,
+    float* __restrict__ part,  // [gridDim.x][cols]
+    int64_t rows, int64_t cols) {
+  const int64_t quads = cols >> 2;
+  float acc[DB_MAX_K][4];
+#pragma unroll
+  for (int k = 0; k < DB_MAX_K; ++k)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[k][j] = 0.f;
+  const int64_t row_stride = (int64_t)gridDim.x * DB_ROWS;
+  for (int64_t r0 = (int64_t)blockIdx.x * DB_ROWS; r0 < rows;
+       r0 += row_stride) {
+    const int nr = min((int64_t)DB_ROWS, rows - r0);
+    float w[DB_ROWS];
+#pragma unroll
+    for (int rr = 0; rr < DB_ROWS; ++rr)
+      w[rr] = rr < nr ? Io::load(dy, r0 + rr) : 0.f;
+    int k = 0;
+    for (int64_t q = threadIdx.x; q < quads; q += blockDim.x, ++k) {
+      float v[DB_ROWS][4];
+#pragma unroll
+      for (int rr = 0; rr < DB_ROWS; ++rr)
+        if (rr < nr) QuadIo<Io>::load4(x, (r0 + rr) * quads + q, v[rr]);
+#pragma unroll
+      for (int rr = 0; rr < DB_ROWS; ++rr)
+        if (rr < nr)
+#pragma unroll
+          for (int j = 0; j < 4; ++j) acc[k][j] += w[rr] * v[rr][j];
+    }
+  }
+  float* prow = part + (int64_t)blockIdx.x * cols;
+  int k = 0;
+  for (int64_t q = threadIdx.x; q < quads; q += blockDim.x, ++k) {
+    f32x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o[j] = acc[k][j];
+    reinterpret_cast<f32x4*>(prow)[q] = o;
+  }
+}
+
+}  // namespace
+
+torch::Tensor col_reduce_dot(torch::Tensor x, torch::Tensor dy) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 2,
+              "x must be contiguous 2D on GPU");
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() &&
+              dy.numel() == x.size(0), "dy must be [rows]");
+  TORCH_CHECK(dy.scalar_type() == x.scalar_type(), "dtype mismatch");
+  const int64_t rows = x.size(0);
+  const int64_t cols = x.size(1);
+  TORCH_CHECK(cols % 4 == 0 && cols <= 4 * MIYARN_BLOCK * DB_MAX_K,
+              "col_reduce_dot needs cols % 4 == 0 and small cols");
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int64_t quads = cols / 4;
+  int block = static_cast<int>(
+      std::min<int64_t>(MIYARN_BLOCK, ((quads + 63) / 64) * 64));
+  int grid = static_cast<int>(std::min<int64_t>(
+      (rows + DB_ROWS - 1) / DB_ROWS, MIYARN_MAX_BLOCKS));
+  auto part = torch::empty({grid, cols},
+                           x.options().dtype(torch::kFloat32));
+  if (x.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(col_reduce_dot_kernel<F32Io>, dim3(grid),
+                       dim3(block), 0, stream, x.data_ptr<float>(),
+                       dy.data_ptr<float>(), part.data_ptr<float>(),
+                       rows, cols);
+  } else {
+    TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "fp32/bf16 only");
+    hipLaunchKernelGGL(col_reduce_dot_kernel<Bf16Io>, dim3(grid),
+                       dim3(block), 0, stream,
+                       reinterpret_cast<unsigned short*>(x.data_ptr()),
+                       reinterpret_cast<unsigned short*>(dy.data_ptr()),
+                       part.data_ptr<float>(), rows, cols);
+  }
+  return part.sum(0);
+}
