@@ -63,6 +63,30 @@ def main():
               f"({gf/t_chunk*1e3:5.0f} TF)")
 
 
+def custom_wgrad():
+    import tf_yarn_amd.ops._C as C
+    for (cin, cout) in LAYERS_PAD:
+        x = torch.randn(B, cin, device="cuda").to(torch.bfloat16)
+        dy = torch.randn(B, cout, device="cuda").to(torch.bfloat16)
+        for sk in (2, 4, 8, 16):
+            t = timeit(lambda: C.wgrad_nt(dy, x, sk))
+            gf = 2 * B * cin * cout / 1e9
+            print(f"[custom wgrad {cout}x{cin} sk={sk:2}] {t:7.1f}us "
+                  f"({gf/t*1e3:5.0f} TF)")
+    x = torch.randn(B, 256, device="cuda").to(torch.bfloat16)
+    dy = torch.randn(B, device="cuda").to(torch.bfloat16)
+    t = timeit(lambda: C.col_reduce_dot(x, dy))
+    print(f"[head col_reduce_dot 256] {t:7.1f}us")
+
+
+LAYERS_PAD = [(448, 1024), (1024, 512), (512, 256)]
+
 if __name__ == "__main__":
     assert torch.cuda.is_available()
-    main()
+    import sys as _sys
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    if "--custom" in _sys.argv:
+        custom_wgrad()
+    else:
+        main()
+        custom_wgrad()

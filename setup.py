@@ -22,6 +22,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "fused_optimizers.hip"),
         os.path.join(CSRC, "embedding.hip"),
         os.path.join(CSRC, "elementwise.hip"),
+        os.path.join(CSRC, "wgrad.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3"],

@@ -358,3 +358,31 @@ def test_bias_relu_fwd_vec_gpu():
         y = ops.bias_relu_fwd(x, b)
         ref = torch.relu(x.float() + b.float()).to(dtype)
         assert torch.allclose(y.float(), ref.float(), atol=1e-2)
+
+
+@requires_gpu
+@pytest.mark.parametrize("B,N,M", [(4096, 128, 64), (8192, 1024, 448),
+                                   (65536, 256, 512)])
+def test_wgrad_nt_matches_reference(B, N, M):
+    import tf_yarn_amd.ops._C as C
+    torch.manual_seed(15)
+    dy = (torch.randn(B, N, device="cuda") / 8).to(torch.bfloat16)
+    x = (torch.randn(B, M, device="cuda") / 8).to(torch.bfloat16)
+    out = C.wgrad_nt(dy, x, 0)
+    ref = dy.float().t().mm(x.float())
+    # bf16 inputs, fp32 accumulate both sides; tolerance scales with sqrt(K)
+    assert out.shape == (N, M)
+    err = (out - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err < 0.02 * max(1.0, scale), f"max err {err} scale {scale}"
+
+
+@requires_gpu
+def test_wgrad_nt_splitk_variants_agree():
+    import tf_yarn_amd.ops._C as C
+    torch.manual_seed(16)
+    dy = (torch.randn(8192, 64, device="cuda") / 8).to(torch.bfloat16)
+    x = (torch.randn(8192, 64, device="cuda") / 8).to(torch.bfloat16)
+    a = C.wgrad_nt(dy, x, 1)
+    b = C.wgrad_nt(dy, x, 8)
+    assert torch.allclose(a, b, atol=1e-2, rtol=1e-3)

@@ -50,6 +50,9 @@ torch::Tensor bias_relu_bwd(torch::Tensor dy, torch::Tensor y);
 std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
                                             torch::Tensor y);
 torch::Tensor col_reduce_dot(torch::Tensor x, torch::Tensor dy);
+
+// wgrad.hip
+torch::Tensor wgrad_nt(torch::Tensor dy, torch::Tensor x, int64_t splitk);
 void convert_scaled(torch::Tensor src, torch::Tensor dst, double scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -67,6 +70,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Sparse embedding grad scatter into dense grad table");
   m.def("bias_relu_fwd", &bias_relu_fwd, "Fused bias+ReLU forward");
   m.def("bias_relu_bwd", &bias_relu_bwd, "Fused ReLU backward");
+  m.def("wgrad_nt", &wgrad_nt,
+        "Split-K MFMA weight gradient: dW = dy^T @ x (bf16 in, fp32 out)");
   m.def("col_reduce_dot", &col_reduce_dot,
         "dw[m] = sum_b dy[b] * x[b,m] (single-logit head wgrad)");
   m.def("bias_relu_bwd_db", &bias_relu_bwd_db,
