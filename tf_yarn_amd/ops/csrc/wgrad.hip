@@ -26,11 +26,18 @@ namespace {
 
 #define WG_BN 64
 #define WG_BM 64
-#define WG_BK 32
-#define BKPAD 40  // LDS k-stride (elements): 80 B rows, conflict-free b128
+#define WG_BK 64
+// LDS k-stride (elements): 136 B rows -> transposed b64 writes and b128
+// fragment reads both bank-conflict-free (banking analysis in module doc)
+#define BKPAD 68
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_v;
 typedef __attribute__((ext_vector_type(4))) float f32x4_v;
+
+union U16x4 {
+  unsigned short u[4];
+  unsigned long long ll;
+};
 
 __global__ __launch_bounds__(256)
 void wgrad_nt_kernel(const unsigned short* __restrict__ dy,
@@ -55,10 +62,17 @@ void wgrad_nt_kernel(const unsigned short* __restrict__ dy,
   const int wn = (wave >> 1) * 32;  // wave quadrant in the 64x64 tile
   const int wm = (wave & 1) * 32;
 
-  // staging map: thread t loads 8 bf16 (16 B) of row r = t/8,
-  // columns c8*8..c8*8+7, then writes them transposed (8 ds_write_u16)
-  const int st_r = tid >> 3;        // 0..31 (k within the BK step)
-  const int st_c = (tid & 7) * 8;   // 0..56 (n/m within the tile)
+  // staging: threads 0-127 stage dy, 128-255 stage x.  Each thread loads
+  // 4 k-rows x 8 columns (4 x 16 B coalesced) and writes 8 packed
+  // ds_write_b64 (4 k-values per write) into the transposed image.
+  const int st_tile = tid >> 7;          // 0 = dy, 1 = x
+  const int st_t = tid & 127;
+  const int st_kg = (st_t >> 3) * 4;     // k base: 0,4,...,60
+  const int st_c = (st_t & 7) * 8;       // col base: 0,8,...,56
+  const unsigned short* st_src = st_tile ? x : dy;
+  unsigned short* st_dst = st_tile ? xT : dyT;
+  const int st_ld = st_tile ? M : N;
+  const int st_o0 = st_tile ? m0 : n0;
 
   f32x4_v acc[2][2];
 #pragma unroll
@@ -70,43 +84,50 @@ void wgrad_nt_kernel(const unsigned short* __restrict__ dy,
   const int a_k = (lane >> 4) * 8;    // fragment k base
 
   for (int64_t k0 = k_begin; k0 < k_end; k0 += WG_BK) {
-    // ---- stage transposed ------------------------------------------------
+    // ---- stage transposed (packed b64 writes) ---------------------------
     {
       const unsigned short* src =
-          dy + (k0 + st_r) * (int64_t)N + n0 + st_c;
-      bf16x4 v0 = reinterpret_cast<const bf16x4*>(src)[0];
-      bf16x4 v1 = reinterpret_cast<const bf16x4*>(src)[1];
+          st_src + (k0 + st_kg) * (int64_t)st_ld + st_o0 + st_c;
+      bf16x4 r0a = reinterpret_cast<const bf16x4*>(src)[0];
+      bf16x4 r0b = reinterpret_cast<const bf16x4*>(src)[1];
+      src += st_ld;
+      bf16x4 r1a = reinterpret_cast<const bf16x4*>(src)[0];
+      bf16x4 r1b = reinterpret_cast<const bf16x4*>(src)[1];
+      src += st_ld;
+      bf16x4 r2a = reinterpret_cast<const bf16x4*>(src)[0];
+      bf16x4 r2b = reinterpret_cast<const bf16x4*>(src)[1];
+      src += st_ld;
+      bf16x4 r3a = reinterpret_cast<const bf16x4*>(src)[0];
+      bf16x4 r3b = reinterpret_cast<const bf16x4*>(src)[1];
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
-        dyT[(st_c + j) * BKPAD + st_r] = v0[j];
-        dyT[(st_c + 4 + j) * BKPAD + st_r] = v1[j];
-      }
-      const unsigned short* srcx =
-          x + (k0 + st_r) * (int64_t)M + m0 + st_c;
-      bf16x4 w0 = reinterpret_cast<const bf16x4*>(srcx)[0];
-      bf16x4 w1 = reinterpret_cast<const bf16x4*>(srcx)[1];
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        xT[(st_c + j) * BKPAD + st_r] = w0[j];
-        xT[(st_c + 4 + j) * BKPAD + st_r] = w1[j];
+        U16x4 pa{{r0a[j], r1a[j], r2a[j], r3a[j]}};
+        *reinterpret_cast<unsigned long long*>(
+            st_dst + (st_c + j) * BKPAD + st_kg) = pa.ll;
+        U16x4 pb{{r0b[j], r1b[j], r2b[j], r3b[j]}};
+        *reinterpret_cast<unsigned long long*>(
+            st_dst + (st_c + 4 + j) * BKPAD + st_kg) = pb.ll;
       }
     }
     __syncthreads();
-    // ---- MFMA ------------------------------------------------------------
-    bf16x8_v a[2], b[2];
+    // ---- MFMA: 2x2 fragments x 2 k-halves -------------------------------
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      a[i] = *reinterpret_cast<const bf16x8_v*>(
-          dyT + (wn + i * 16 + a_row) * BKPAD + a_k);
-      b[i] = *reinterpret_cast<const bf16x8_v*>(
-          xT + (wm + i * 16 + a_row) * BKPAD + a_k);
+    for (int kh = 0; kh < 2; ++kh) {
+      bf16x8_v a[2], b[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        a[i] = *reinterpret_cast<const bf16x8_v*>(
+            dyT + (wn + i * 16 + a_row) * BKPAD + kh * 32 + a_k);
+        b[i] = *reinterpret_cast<const bf16x8_v*>(
+            xT + (wm + i * 16 + a_row) * BKPAD + kh * 32 + a_k);
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[i], b[j], acc[i][j], 0, 0, 0);
     }
-#pragma unroll
-    for (int i = 0; i < 2; ++i)
-#pragma unroll
-      for (int j = 0; j < 2; ++j)
-        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a[i], b[j], acc[i][j], 0, 0, 0);
     __syncthreads();
   }
 
