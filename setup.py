@@ -10,10 +10,21 @@ import os
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
+from pybind11.setup_helpers import Pybind11Extension  # noqa: E402
 from setuptools import find_packages, setup  # noqa: E402
-from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+from torch.utils.cpp_extension import (BuildExtension,  # noqa: E402
+                                       CUDAExtension)
 
 CSRC = os.path.join("tf_yarn_amd", "ops", "csrc")
+
+# torch-free (imports without loading libtorch; the control plane must be
+# usable in any task process)
+kv_ext = Pybind11Extension(
+    "tf_yarn_amd._kv_native",
+    sources=[os.path.join("tf_yarn_amd", "csrc_kv", "kv_server.cpp")],
+    cxx_std=17,
+    extra_compile_args=["-O2", "-pthread"],
+)
 
 ext = CUDAExtension(
     name="tf_yarn_amd.ops._C",
@@ -36,6 +47,6 @@ setup(
     description=("MI355X-native distributed-training launcher with "
                  "criteo/tf-yarn's capabilities"),
     packages=find_packages(include=["tf_yarn_amd", "tf_yarn_amd.*"]),
-    ext_modules=[ext],
+    ext_modules=[ext, kv_ext],
     cmdclass={"build_ext": BuildExtension},
 )

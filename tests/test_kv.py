@@ -128,3 +128,39 @@ def test_large_value_roundtrip(kv_client):
     blob = bytes(range(256)) * (4 * 1024 * 16)  # 4 MiB
     kv_client.put("blob", blob)
     assert kv_client.get("blob") == blob
+
+
+def test_python_fallback_server_same_protocol():
+    """The pure-Python server must speak the identical wire protocol."""
+    from tf_yarn_amd.kv import PyKVServer
+    server = PyKVServer()
+    try:
+        c = KVClient(server.address)
+        c.put("a", b"1")
+        assert c.get("a") == b"1"
+        assert c.add("n", 3) == 3
+        assert c.compare_set("z", b"", b"v") == b"v"
+        assert c.compare_set("z", b"wrong", b"x") == b"v"
+        assert c.list("") == {"a": b"1", "n": b"3", "z": b"v"}
+        with pytest.raises(TimeoutError):
+            c.wait("missing", timeout=0.1)
+    finally:
+        server.stop()
+
+
+def test_native_server_selected_when_built():
+    from tf_yarn_amd import kv
+    if kv._native_available():
+        assert isinstance(kv.KVServer(), kv.NativeKVServer) or True
+        s = kv.KVServer()
+        assert type(s).__name__ == "NativeKVServer"
+        s.stop()
+
+
+def test_add_and_cas(kv_client):
+    assert kv_client.add("counter", 10) == 10
+    assert kv_client.add("counter", -3) == 7
+    assert kv_client.get("counter") == b"7"
+    assert kv_client.compare_set("k", b"", b"a") == b"a"
+    assert kv_client.compare_set("k", b"a", b"b") == b"b"
+    assert kv_client.compare_set("k", b"a", b"c") == b"b"
