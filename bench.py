@@ -91,10 +91,14 @@ def main() -> None:
                      else torch.float32)
     table_sizes = [args.table_rows] * 26
 
-    torch.manual_seed(1234)  # identical random-init across ranks
+    torch.manual_seed(1234 + rank)  # sharded tables are per-rank
+    # Feature-sharded embeddings + all-to-all: the xGMI-native scaling
+    # path (replicated tables would allgather ~54 MB/rank/step).  Dense
+    # params still random-init identically via the reducer's broadcast.
     model = WideAndDeep(table_sizes=table_sizes,
                         embedding_dim=EMBEDDING_DIM, hidden=HIDDEN,
-                        compute_dtype=compute_dtype).to(device)
+                        compute_dtype=compute_dtype,
+                        sharded=True).to(device)
     ddp = BucketedDataParallel(model) if world_size > 1 else model
     module = ddp.module if world_size > 1 else model
 

@@ -323,16 +323,30 @@ class WideAndDeep(nn.Module):
                  table_sizes: Optional[List[int]] = None,
                  embedding_dim: int = 16,
                  hidden: Tuple[int, ...] = (1024, 512, 256),
-                 compute_dtype: torch.dtype = torch.float32):
+                 compute_dtype: torch.dtype = torch.float32,
+                 sharded: bool = False,
+                 process_group=None):
         super().__init__()
         table_sizes = table_sizes or DEFAULT_TABLE_SIZES
         self.compute_dtype = compute_dtype
+        self.dense_dim = dense_dim
         bf16 = compute_dtype == torch.bfloat16
-        self.deep_embedding = SparseEmbedding(table_sizes, embedding_dim,
-                                              out_bf16=bf16)
-        # Wide part: one scalar weight per categorical id (linear-in-one-hot)
-        self.wide_embedding = WideScalarEmbedding(table_sizes,
-                                                  out_bf16=bf16)
+        self.sharded = sharded
+        if sharded:
+            from tf_yarn_amd.models.sharded_embedding import \
+                ShardedCriteoEmbeddings
+            self.embeddings = ShardedCriteoEmbeddings(
+                table_sizes, embedding_dim, out_bf16=bf16,
+                process_group=process_group)
+            self.deep_embedding = None
+            self.wide_embedding = None
+        else:
+            self.embeddings = None
+            self.deep_embedding = SparseEmbedding(
+                table_sizes, embedding_dim, out_bf16=bf16)
+            # Wide: one scalar weight per categorical id
+            self.wide_embedding = WideScalarEmbedding(table_sizes,
+                                                      out_bf16=bf16)
         self.wide_dense = ScalarHead(dense_dim, compute_dtype)
         layers: List[nn.Module] = []
         # dense block padded to 16 columns for quad-aligned fused gather
@@ -349,6 +363,19 @@ class WideAndDeep(nn.Module):
     def forward(self, dense: torch.Tensor,
                 sparse_ids: torch.Tensor) -> torch.Tensor:
         dense = dense.to(self.compute_dtype)
+        if self.sharded:
+            b = dense.shape[0]
+            emb = self.embeddings
+            deep_in = torch.empty(
+                b, _DeepInput.DENSE_PAD + emb.F * emb.dim,
+                dtype=self.compute_dtype, device=dense.device)
+            deep_in[:, :self.dense_dim] = dense
+            deep_in[:, self.dense_dim:_DeepInput.DENSE_PAD] = 0
+            deep_in, wide_sum = emb(sparse_ids, deep_in,
+                                    _DeepInput.DENSE_PAD)
+            deep_out = self.head(self.mlp(deep_in))
+            return deep_out + wide_sum.to(deep_out.dtype) \
+                + self.wide_dense(dense)
         emb = self.deep_embedding
         flat = (sparse_ids + emb.offsets.unsqueeze(0)).reshape(-1)
         deep_in = _DeepInput.apply(
@@ -359,13 +386,19 @@ class WideAndDeep(nn.Module):
         return deep_out + wide_out
 
     def start_sparse_sync(self, process_group=None) -> None:
-        """Kick off both embeddings' allgathers (call right after
-        ``loss.backward()`` so the communication overlaps with the dense
-        optimizer step)."""
+        """Kick off the replicated-mode allgathers (called right after
+        ``loss.backward()`` so communication overlaps with the dense
+        optimizer step).  Sharded mode already exchanged grads during
+        backward (all-to-all), so there is nothing to start."""
+        if self.sharded:
+            return
         self.deep_embedding.start_sparse_sync(process_group)
         self.wide_embedding.start_sparse_sync(process_group)
 
     def finish_sparse_sync(self, lr: float) -> None:
+        if self.sharded:
+            self.embeddings.apply_sparse_updates(lr)
+            return
         self.deep_embedding.finish_sparse_sync(lr)
         self.wide_embedding.finish_sparse_sync(lr)
 
@@ -374,5 +407,8 @@ class WideAndDeep(nn.Module):
         self.finish_sparse_sync(lr)
 
     def clear_pending(self) -> None:
+        if self.sharded:
+            self.embeddings.clear_pending()
+            return
         self.deep_embedding.clear_pending()
         self.wide_embedding.clear_pending()

@@ -164,7 +164,10 @@ class BucketedDataParallel(nn.Module):
             for buf in self.module.buffers():
                 dist.broadcast(buf, src=0, group=self.process_group)
             for p in self.module.parameters():
-                if getattr(p, "_miyarn_sparse", False):
+                if getattr(p, "_miyarn_sparse", False) \
+                        and not getattr(p, "_miyarn_sharded", False):
+                    # replicated sparse tables start rank-0-equal;
+                    # sharded tables are per-rank by construction
                     dist.broadcast(p.detach(), src=0,
                                    group=self.process_group)
         # Grads for non-bucketed params (requires_grad=False) never sync.
