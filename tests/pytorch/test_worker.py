@@ -52,3 +52,32 @@ def test_create_dataloader_iterable_passthrough():
         rank=0, world_size=2)
     assert loader.sampler is None or not isinstance(
         loader.sampler, torch.utils.data.distributed.DistributedSampler)
+
+
+def test_parquet_dataset_sharding(tmp_path):
+    """ParquetDataset: per-rank contiguous batch slices, equal counts,
+    ragged tail dropped (reference parquet_dataset.py semantics)."""
+    pa = pytest.importorskip("pyarrow")
+    import pyarrow.parquet as pq
+
+    from tf_yarn_amd.pytorch.parquet_dataset import ParquetDataset
+
+    n_rows = 103  # ragged: 103 rows, batch 10 -> 10 full batches
+    table = pa.table({"x": list(range(n_rows))})
+    path = str(tmp_path / "part-0.parquet")
+    pq.write_table(table, path)
+
+    ds = ParquetDataset([path], batch_size=10)
+    batches = list(ds)
+    assert len(batches) == 10  # single shard gets all full batches
+    assert all(b.num_rows == 10 for b in batches)
+    seen = [v for b in batches for v in b.column("x").to_pylist()]
+    assert seen == list(range(100))  # ragged tail (3 rows) dropped
+
+    # two-shard split: equal counts, contiguous, disjoint
+    ds.world_size, ds.rank = 2, 0
+    first = [v for b in ds for v in b.column("x").to_pylist()]
+    ds.rank = 1
+    second = [v for b in ds for v in b.column("x").to_pylist()]
+    assert len(first) == len(second) == 50
+    assert set(first).isdisjoint(second)

@@ -16,7 +16,6 @@ from __future__ import annotations
 import logging
 import os
 import re
-import time
 from typing import Callable, Dict, List, NamedTuple, Optional, Sequence
 
 import torch

@@ -12,7 +12,7 @@ from __future__ import annotations
 
 import logging
 import os
-from typing import Callable, Dict, List, Optional, Sequence, Union
+from typing import Callable, Dict, List, Optional, Sequence
 
 import torch
 from torch import nn

@@ -15,8 +15,6 @@ from __future__ import annotations
 import logging
 import os
 import sys
-from typing import Optional, Union
-
 import torch
 
 from tf_yarn_amd import _task_commons, event

@@ -8,7 +8,7 @@ bucketed-allreduce reducer.
 
 from __future__ import annotations
 
-from typing import List, Type
+from typing import List
 
 import torch
 from torch import nn

@@ -11,8 +11,7 @@ from typing import List, Optional, Tuple
 
 import torch
 
-from tf_yarn_amd.models.wide_deep import (CRITEO_DENSE, CRITEO_SPARSE,
-                                          DEFAULT_TABLE_SIZES)
+from tf_yarn_amd.models.wide_deep import CRITEO_DENSE, DEFAULT_TABLE_SIZES
 
 
 def synthetic_criteo_batch(batch_size: int,

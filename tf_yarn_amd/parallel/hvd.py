@@ -12,7 +12,7 @@ same bucket buffers.
 from __future__ import annotations
 
 import logging
-from typing import Iterable, List, Optional
+from typing import List
 
 import torch
 import torch.distributed as dist
