@@ -106,3 +106,16 @@ def test_train_and_evaluate(tmp_path):
                        EvalSpec(_input_fn, steps=2))
     assert est.global_step == 5
     assert evaluated_steps(est.eval_dir()) == [5]
+
+
+def test_get_all_metrics(tmp_path):
+    from tf_yarn_amd.estimator.metrics import get_all_metrics
+    from tf_yarn_amd.utils import tb
+    w = tb.SummaryWriter(str(tmp_path))
+    w.add_scalar("loss", 0.5, step=1)
+    w.add_scalar("loss", 0.25, step=2)
+    w.close()
+    m = get_all_metrics(str(tmp_path))
+    assert m["name"] == ["loss", "loss"]
+    assert m["value"] == [0.5, 0.25]
+    assert m["step"] == [1, 2]

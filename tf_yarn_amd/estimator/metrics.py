@@ -102,3 +102,18 @@ def _add_monitor_to_experiment(
     # hooks are passed at train time by the task modules; nothing to
     # mutate on the descriptor itself in the torch-backed design
     return experiment
+
+
+def get_all_metrics(model_dir: str):
+    """Parse the event files under *model_dir* into
+    {"step": [...], "name": [...], "value": [...]} (reference
+    ``tensorflow/metrics.py:74-108`` parses TB event files the same way)."""
+    from tf_yarn_amd.utils import tb
+    steps, names, values = [], [], []
+    for ev in tb.read_events(model_dir):
+        if ev.get("value") is None:
+            continue
+        steps.append(ev.get("step"))
+        names.append(ev.get("tag"))
+        values.append(ev.get("value"))
+    return {"step": steps, "name": names, "value": values}
