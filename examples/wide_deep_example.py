@@ -1,5 +1,5 @@
 """Flagship example: Criteo wide-and-deep CTR training across the node's
-GPUs (BASELINE config 3 shape) — fused embedding gather/scatter HIP
+GPUs (BASELINE config 3 shape) - fused embedding gather/scatter HIP
 kernels, bf16 compute, bucketed ring allreduce for the dense MLP, sparse
 allgather sync for the embeddings.
 

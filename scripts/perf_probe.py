@@ -10,6 +10,18 @@ import os
 import sys
 import time
 
+_REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+_TUNED = os.path.join(_REPO, "tuned", "tunableop_wide_deep.csv")
+os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+if os.path.exists(_TUNED) and "PYTORCH_TUNABLEOP_FILENAME" not in os.environ:
+    import shutil
+    import tempfile
+    _tdir = tempfile.mkdtemp(prefix="miyarn_tuned_")
+    shutil.copy(_TUNED, os.path.join(_tdir, "probe_tuned0.csv"))
+    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = os.path.join(
+        _tdir, "probe_tuned.csv")
+
 import torch
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
@@ -26,7 +38,7 @@ def probe(batch: int, table_rows: int, steps: int = 20,
     torch.manual_seed(0)
     model = WideAndDeep(table_sizes=tables, embedding_dim=16,
                         hidden=(1024, 512, 256),
-                        compute_dtype=dtype).to(device)
+                        compute_dtype=dtype, sharded=True).to(device)
     opt = FusedSGD([p for p in model.parameters()
                     if not getattr(p, "_miyarn_sparse", False)], lr=0.02)
     loss_fn = torch.nn.BCEWithLogitsLoss()
