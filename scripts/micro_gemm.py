@@ -73,6 +73,14 @@ def custom_wgrad():
             gf = 2 * B * cin * cout / 1e9
             print(f"[custom wgrad {cout}x{cin} sk={sk:2}] {t:7.1f}us "
                   f"({gf/t*1e3:5.0f} TF)")
+    for (cin, cout) in [(432, 1024)] + LAYERS_PAD[1:]:
+        x = torch.randn(B, cin, device="cuda").to(torch.bfloat16)
+        dy = torch.randn(B, cout, device="cuda").to(torch.bfloat16)
+        for sk in (4, 8, 16, 32):
+            t = timeit(lambda: C.wgrad_nt128(dy, x, sk))
+            gf = 2 * B * cin * cout / 1e9
+            print(f"[wgrad128 {cout}x{cin} sk={sk:2}] {t:7.1f}us "
+                  f"({gf/t*1e3:5.0f} TF)")
     x = torch.randn(B, 256, device="cuda").to(torch.bfloat16)
     dy = torch.randn(B, device="cuda").to(torch.bfloat16)
     t = timeit(lambda: C.col_reduce_dot(x, dy))

@@ -34,6 +34,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "embedding.hip"),
         os.path.join(CSRC, "elementwise.hip"),
         os.path.join(CSRC, "wgrad.hip"),
+        os.path.join(CSRC, "wgrad128.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3"],

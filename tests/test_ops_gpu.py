@@ -386,3 +386,19 @@ def test_wgrad_nt_splitk_variants_agree():
     a = C.wgrad_nt(dy, x, 1)
     b = C.wgrad_nt(dy, x, 8)
     assert torch.allclose(a, b, atol=1e-2, rtol=1e-3)
+
+
+@requires_gpu
+@pytest.mark.parametrize("B,N,M", [(4096, 128, 64), (8192, 1024, 432),
+                                   (65536, 256, 512), (4096, 128, 136)])
+def test_wgrad_nt128_matches_reference(B, N, M):
+    import tf_yarn_amd.ops._C as C
+    torch.manual_seed(17)
+    dy = (torch.randn(B, N, device="cuda") / 8).to(torch.bfloat16)
+    x = (torch.randn(B, M, device="cuda") / 8).to(torch.bfloat16)
+    out = C.wgrad_nt128(dy, x, 0)
+    ref = dy.float().t().mm(x.float())
+    assert out.shape == (N, M)
+    err = (out - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err < 0.02 * max(1.0, scale), f"max err {err} scale {scale}"
