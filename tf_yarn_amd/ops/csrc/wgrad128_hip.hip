@@ -79,53 +79,53 @@ void wgrad128_kernel(const unsigned short* __restrict__ dy,
   const int a_row = lane & 15;
   const int a_k = (lane >> 4) * 8;
   const bool m_edge = (m0 + W3_BM) > M;
+  const bool in0 = !m_edge || (m0 + st_c + 4) <= M;
+  const bool in1 = !m_edge || (m0 + st_c + 8) <= M;
 
+  // T14 split (guide G15): issue the NEXT step's global loads before this
+  // step's MFMA phase so HBM latency hides under compute; the register
+  // tile is written to LDS after the barrier.
+  bf16x4 rdy[4][2], rx[4][2];
+
+  auto load_step = [&](int64_t k0) {
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const unsigned short* src =
+          dy + (k0 + st_kg + rr) * (int64_t)N + n0 + st_c;
+      rdy[rr][0] = reinterpret_cast<const bf16x4*>(src)[0];
+      rdy[rr][1] = reinterpret_cast<const bf16x4*>(src)[1];
+      const unsigned short* srcx =
+          x + (k0 + st_kg + rr) * (int64_t)M + m0 + st_c;
+      // real branches (EXEC-masked loads): an out-of-range lane must not
+      // issue the load at all (last row would read past the tensor)
+      rx[rr][0] = bf16x4{0, 0, 0, 0};
+      rx[rr][1] = bf16x4{0, 0, 0, 0};
+      if (in0) rx[rr][0] = reinterpret_cast<const bf16x4*>(srcx)[0];
+      if (in1) rx[rr][1] = reinterpret_cast<const bf16x4*>(srcx)[1];
+    }
+  };
+
+  auto write_step = [&] {
+#pragma unroll
+    for (int h = 0; h < 2; ++h)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        U16x4b p{{rdy[0][h][j], rdy[1][h][j], rdy[2][h][j],
+                  rdy[3][h][j]}};
+        *reinterpret_cast<unsigned long long*>(
+            dyT + lds_off(st_c + h * 4 + j, st_kg)) = p.ll;
+        U16x4b q{{rx[0][h][j], rx[1][h][j], rx[2][h][j], rx[3][h][j]}};
+        *reinterpret_cast<unsigned long long*>(
+            xT + lds_off(st_c + h * 4 + j, st_kg)) = q.ll;
+      }
+  };
+
+  load_step(k_begin);
   for (int64_t k0 = k_begin; k0 < k_end; k0 += W3_BK) {
-    // ---- stage dy ([64k][128n] slice -> [n][k] swizzled image) ----------
-    {
-      bf16x4 r[4][2];
-#pragma unroll
-      for (int rr = 0; rr < 4; ++rr) {
-        const unsigned short* src =
-            dy + (k0 + st_kg + rr) * (int64_t)N + n0 + st_c;
-        r[rr][0] = reinterpret_cast<const bf16x4*>(src)[0];
-        r[rr][1] = reinterpret_cast<const bf16x4*>(src)[1];
-      }
-#pragma unroll
-      for (int h = 0; h < 2; ++h)
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          U16x4b p{{r[0][h][j], r[1][h][j], r[2][h][j], r[3][h][j]}};
-          *reinterpret_cast<unsigned long long*>(
-              dyT + lds_off(st_c + h * 4 + j, st_kg)) = p.ll;
-        }
-    }
-    // ---- stage x (same pattern, column-masked at the M edge) ------------
-    {
-      bf16x4 r[4][2];
-      const bool in0 = !m_edge || (m0 + st_c + 4) <= M;
-      const bool in1 = !m_edge || (m0 + st_c + 8) <= M;
-#pragma unroll
-      for (int rr = 0; rr < 4; ++rr) {
-        const unsigned short* src =
-            x + (k0 + st_kg + rr) * (int64_t)M + m0 + st_c;
-        // real branches (EXEC-masked loads): an out-of-range lane must
-        // not issue the load at all (last row would read past the tensor)
-        r[rr][0] = bf16x4{0, 0, 0, 0};
-        r[rr][1] = bf16x4{0, 0, 0, 0};
-        if (in0) r[rr][0] = reinterpret_cast<const bf16x4*>(src)[0];
-        if (in1) r[rr][1] = reinterpret_cast<const bf16x4*>(src)[1];
-      }
-#pragma unroll
-      for (int h = 0; h < 2; ++h)
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          U16x4b p{{r[0][h][j], r[1][h][j], r[2][h][j], r[3][h][j]}};
-          *reinterpret_cast<unsigned long long*>(
-              xT + lds_off(st_c + h * 4 + j, st_kg)) = p.ll;
-        }
-    }
+    write_step();
     __syncthreads();
+    if (k0 + W3_BK < k_end)
+      load_step(k0 + W3_BK);  // in flight under the MFMA phase
     // ---- MFMA: 4x4 fragments x 2 k-halves -------------------------------
 #pragma unroll
     for (int kh = 0; kh < 2; ++kh) {
