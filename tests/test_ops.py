@@ -125,3 +125,20 @@ def test_scalar_head_fn_matches_autograd():
     assert torch.allclose(x.grad, x2.grad, atol=1e-5)
     assert torch.allclose(w.grad, w2.grad, atol=1e-4)
     assert torch.allclose(b.grad, b2.grad, atol=1e-5)
+
+
+def test_linear_bias_relu_matches_autograd():
+    torch.manual_seed(5)
+    x = torch.randn(16, 8, requires_grad=True)
+    w = torch.randn(6, 8, requires_grad=True)
+    b = torch.randn(6, requires_grad=True)
+    y = ops.linear_bias_relu(x, w, b)
+    y.pow(2).sum().backward()
+    x2 = x.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    torch.relu(x2 @ w2.t() + b2).pow(2).sum().backward()
+    assert torch.allclose(y, torch.relu(x @ w.t() + b).detach())
+    assert torch.allclose(x.grad, x2.grad, atol=1e-5)
+    assert torch.allclose(w.grad, w2.grad, atol=1e-5)
+    assert torch.allclose(b.grad, b2.grad, atol=1e-5)

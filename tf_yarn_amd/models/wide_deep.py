@@ -311,8 +311,7 @@ class FusedLinearReLU(nn.Module):
         nn.init.kaiming_uniform_(self.weight, a=math.sqrt(5))
 
     def forward(self, x):
-        y = x.matmul(self.weight.t())
-        return ops.bias_relu(y, self.bias)
+        return ops.linear_bias_relu(x, self.weight, self.bias)
 
 
 class WideAndDeep(nn.Module):

@@ -309,3 +309,51 @@ class BiasReLU(torch.autograd.Function):
 
 def bias_relu(x: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
     return BiasReLU.apply(x, bias)
+
+
+def _use_custom_wgrad(dz: torch.Tensor, x: torch.Tensor) -> bool:
+    """Route the weight gradient through the split-K MFMA kernel where it
+    measured faster than hipBLASLt (small outputs: N*M <= 256*512; the
+    library's larger-tile solutions win on the bigger layers)."""
+    if not (dz.is_cuda and HAVE_EXT):
+        return False
+    n, m = dz.size(1), x.size(1)
+    return (dz.dtype == torch.bfloat16 and x.dtype == torch.bfloat16
+            and n % 128 == 0 and m % 8 == 0 and dz.size(0) % 64 == 0
+            and n * m <= 256 * 512)
+
+
+class LinearBiasReLU(torch.autograd.Function):
+    """y = relu(x @ w.T + bias) with a fully-controlled backward:
+    fused dx+dbias kernel, dgrad via hipBLASLt, wgrad via the custom
+    split-K MFMA kernel on shapes where it wins."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        z = x.matmul(weight.t())
+        y = bias_relu_fwd(z, bias)
+        ctx.save_for_backward(x, weight, y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, y = ctx.saved_tensors
+        dy = dy.contiguous()
+        if dy.is_cuda and HAVE_EXT and y.size(-1) % 4 == 0 \
+                and y.size(-1) <= 8192:
+            dz, dbias32 = _C.bias_relu_bwd_db(dy, y.contiguous())
+            dbias = dbias32.to(dy.dtype)
+        else:
+            dz = bias_relu_bwd(dy, y)
+            dbias = dz.sum(dim=tuple(range(dz.dim() - 1)))
+        dx = dz.matmul(weight)
+        if _use_custom_wgrad(dz, x):
+            dw = _C.wgrad_nt128(dz, x, 0).to(weight.dtype)
+        else:
+            dw = dz.t().matmul(x)
+        return dx, dw, dbias
+
+
+def linear_bias_relu(x: torch.Tensor, weight: torch.Tensor,
+                     bias: torch.Tensor) -> torch.Tensor:
+    return LinearBiasReLU.apply(x, weight, bias)
