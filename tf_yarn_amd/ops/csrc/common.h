@@ -14,11 +14,15 @@
 #define MIYARN_BLOCK 256
 #define MIYARN_MAX_BLOCKS 2048  // 256 CUs * 8 blocks/CU
 
-static inline int miyarn_grid(int64_t n_items) {
+static inline int miyarn_grid_cap(int64_t n_items, int64_t cap) {
   int64_t blocks = (n_items + MIYARN_BLOCK - 1) / MIYARN_BLOCK;
-  if (blocks > MIYARN_MAX_BLOCKS) blocks = MIYARN_MAX_BLOCKS;
+  if (blocks > cap) blocks = cap;
   if (blocks < 1) blocks = 1;
   return static_cast<int>(blocks);
+}
+
+static inline int miyarn_grid(int64_t n_items) {
+  return miyarn_grid_cap(n_items, MIYARN_MAX_BLOCKS);
 }
 
 // ---- bf16 <-> f32 ----------------------------------------------------------

@@ -240,7 +240,9 @@ void emb_bwd_sgd(torch::Tensor table, torch::Tensor ids, torch::Tensor grad,
               grad.numel() == n * dim, "grad shape mismatch");
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const bool vec = (dim % 4 == 0);
-  int grid = miyarn_grid(n * (vec ? dim / 4 : dim));
+  // atomic RMW latency is higher than load latency: give the scheduler
+  // 4x the streaming-kernel wave count to hide it
+  int grid = miyarn_grid_cap(n * (vec ? dim / 4 : dim), 8192);
   float nls = (float)(-lr * scale);
   if (grad.scalar_type() == torch::kFloat32) {
     if (vec)
