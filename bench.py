@@ -70,8 +70,11 @@ def main() -> None:
     parser.add_argument("--table-rows", type=int,
                         default=TABLE_ROWS_PER_FEATURE,
                         help="rows per categorical table (tests use small)")
-    parser.add_argument("--no-graphs", action="store_true",
-                        help="disable hipGraph step capture")
+    parser.add_argument("--graphs", action="store_true",
+                        help="capture the step in a hipGraph (measured "
+                             "2.39 vs 2.10 ms/step eager on MI355X — "
+                             "replay sustains higher power and downclocks"
+                             ", so off by default)")
     args = parser.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -134,7 +137,7 @@ def main() -> None:
         return loss
 
     step = eager_step
-    if use_gpu and not args.no_graphs:
+    if use_gpu and args.graphs:
         # hipGraph capture: the step is launch-dense (~60 kernels); one
         # graph per pre-generated batch replays it with zero host gaps.
         # Capture happens after warmup so optimizer state and tuned GEMM
