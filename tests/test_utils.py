@@ -1,0 +1,79 @@
+"""Utils: phase timer, board server, fs seam, summary writer."""
+
+import json
+import time
+import urllib.request
+
+import pytest
+import torch
+
+from tf_yarn_amd import tensorboard
+from tf_yarn_amd.utils import tb, trace
+from tf_yarn_amd.utils.fs import LocalFs, resolve_filesystem_and_path
+
+
+def test_phase_timer_cpu():
+    timer = trace.PhaseTimer(use_cuda=False)
+    for _ in range(3):
+        with timer.phase("work"):
+            time.sleep(0.01)
+    s = timer.summary()
+    assert s["work"]["count"] == 3
+    assert s["work"]["median_ms"] >= 9.0
+
+
+def test_phase_timer_chrome_trace(tmp_path):
+    timer = trace.PhaseTimer(use_cuda=False)
+    with timer.phase("a"):
+        pass
+    path = str(tmp_path / "trace.json")
+    timer.export_chrome_trace(path)
+    data = json.load(open(path))
+    assert data["traceEvents"][0]["name"] == "a"
+
+
+def test_board_server_serves_metrics(tmp_path):
+    writer = tb.SummaryWriter(str(tmp_path))
+    writer.add_scalar("loss", 1.5, step=3)
+    writer.close()
+    server, url = tensorboard.start_tf_board(None, str(tmp_path))
+    try:
+        with urllib.request.urlopen(f"{url}/metrics", timeout=10) as r:
+            events = json.loads(r.read())
+        assert events[0]["tag"] == "loss"
+        assert events[0]["value"] == 1.5
+        with urllib.request.urlopen(url, timeout=10) as r:
+            html = r.read().decode()
+        assert "loss" in html
+    finally:
+        server.shutdown()
+
+
+def test_resolve_filesystem_and_path(tmp_path):
+    fs, path = resolve_filesystem_and_path(f"file://{tmp_path}/x.txt")
+    assert isinstance(fs, LocalFs)
+    assert path == f"{tmp_path}/x.txt"
+    fs2, path2 = resolve_filesystem_and_path("/plain/path")
+    assert path2 == "/plain/path"
+    with pytest.raises(ValueError):
+        resolve_filesystem_and_path("hdfs://nn/path")
+
+
+def test_localfs_roundtrip(tmp_path):
+    fs = LocalFs()
+    src = tmp_path / "a.txt"
+    src.write_text("hello")
+    fs.put(str(src), str(tmp_path / "sub" / "b.txt"))
+    assert fs.exists(str(tmp_path / "sub" / "b.txt"))
+    assert (tmp_path / "sub" / "b.txt").read_text() == "hello"
+    assert str(tmp_path / "a.txt") in fs.ls(str(tmp_path))
+    fs.rm(str(tmp_path / "sub"), recursive=True)
+    assert not fs.exists(str(tmp_path / "sub"))
+
+
+def test_summary_writer_read_events(tmp_path):
+    with tb.SummaryWriter(str(tmp_path)) as w:
+        w.add_scalars("m", {"a": 1.0, "b": 2.0}, step=1)
+    events = tb.read_events(str(tmp_path))
+    tags = {e["tag"] for e in events}
+    assert tags == {"m/a", "m/b"}
