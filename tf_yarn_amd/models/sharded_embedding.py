@@ -45,17 +45,58 @@ def _world_rank(group) -> Tuple[int, int]:
     return 1, 0
 
 
+_ALLTOALL_MODE: dict = {}  # per-group: "native" | "emulate"
+
+
+def _alltoall_self_check(group=None) -> str:
+    """One-time probe: every rank sends rank*100+dest to dest; verify the
+    native all_to_all_single delivers it.  On any mismatch or error the
+    group permanently falls back to the allgather emulation (correct on
+    every backend) — insurance for the unattended 8-GPU scale run."""
+    world = dist.get_world_size(group)
+    rank = dist.get_rank(group)
+    device = (torch.device("cuda", torch.cuda.current_device())
+              if dist.get_backend(group) == "nccl" else "cpu")
+    send = torch.tensor([rank * 100 + d for d in range(world)],
+                        dtype=torch.int64, device=device)
+    recv = torch.zeros(world, dtype=torch.int64, device=device)
+    try:
+        dist.all_to_all_single(recv, send,
+                               output_split_sizes=[1] * world,
+                               input_split_sizes=[1] * world, group=group)
+        expected = torch.tensor([s * 100 + rank for s in range(world)],
+                                dtype=torch.int64, device=device)
+        ok = bool(torch.equal(recv, expected))
+    except Exception:  # noqa: BLE001
+        ok = False
+    # every rank must agree on the mode
+    flag = torch.tensor([1 if ok else 0], dtype=torch.int64,
+                        device=device)
+    dist.all_reduce(flag, op=dist.ReduceOp.MIN, group=group)
+    mode = "native" if int(flag.item()) == 1 else "emulate"
+    if mode == "emulate":
+        logger.warning("all_to_all_single self-check failed; using the "
+                       "allgather emulation for this group")
+    return mode
+
+
 def _all_to_all_single(output: torch.Tensor, input_: torch.Tensor,
                        out_splits: List[int], in_splits: List[int],
                        group=None) -> None:
     """dist.all_to_all_single with a gloo fallback (allgather + slice) so
-    the distributed path is testable on CPU (gloo lacks alltoall)."""
+    the distributed path is testable on CPU (gloo lacks alltoall), plus a
+    one-time native self-check on other backends."""
     backend = dist.get_backend(group)
     if backend != "gloo":
-        dist.all_to_all_single(output, input_,
-                               output_split_sizes=out_splits,
-                               input_split_sizes=in_splits, group=group)
-        return
+        key = id(group)
+        if key not in _ALLTOALL_MODE:
+            _ALLTOALL_MODE[key] = _alltoall_self_check(group)
+        if _ALLTOALL_MODE[key] == "native":
+            dist.all_to_all_single(output, input_,
+                                   output_split_sizes=out_splits,
+                                   input_split_sizes=in_splits,
+                                   group=group)
+            return
     world = dist.get_world_size(group)
     rank = dist.get_rank(group)
     # Emulation: allgather every rank's full input (+ splits), then take
