@@ -63,6 +63,24 @@ def main():
               f"({gf/t_chunk*1e3:5.0f} TF)")
 
 
+def wgrad_b_sweep():
+    """Is wgrad128 bound by redundant HBM traffic or by its inner
+    structure?  At B=8192 both operands fit in L2/L3 -> if the TF rate
+    jumps, traffic is the bound; if flat, the loop structure is."""
+    import tf_yarn_amd.ops._C as C
+    for Bs in (8192, 16384, 65536):
+        dy = torch.randn(Bs, 1024, device="cuda").to(torch.bfloat16)
+        x = torch.randn(Bs, 432, device="cuda").to(torch.bfloat16)
+        gf = 2 * Bs * 1024 * 432 / 1e9
+        t_lib = timeit(lambda: dy.t().matmul(x))
+        best = None
+        for sk in (4, 8, 16, 32):
+            t = timeit(lambda: C.wgrad_nt128(dy, x, sk))
+            best = t if best is None or t < best else best
+        print(f"[B={Bs:6}] lib {t_lib:7.1f}us ({gf/t_lib*1e3:5.0f} TF)  "
+              f"wgrad128 {best:7.1f}us ({gf/best*1e3:5.0f} TF)")
+
+
 def custom_wgrad():
     import tf_yarn_amd.ops._C as C
     for (cin, cout) in LAYERS_PAD:
@@ -93,7 +111,9 @@ if __name__ == "__main__":
     assert torch.cuda.is_available()
     import sys as _sys
     sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
-    if "--custom" in _sys.argv:
+    if "--bsweep" in _sys.argv:
+        wgrad_b_sweep()
+    elif "--custom" in _sys.argv:
         custom_wgrad()
     else:
         main()
