@@ -119,3 +119,42 @@ def test_get_all_metrics(tmp_path):
     assert m["name"] == ["loss", "loss"]
     assert m["value"] == [0.5, 0.25]
     assert m["step"] == [1, 2]
+
+
+def test_exporter_writes_export(tmp_path):
+    from tf_yarn_amd.estimator.estimator import Exporter
+    est = _make_estimator(str(tmp_path / "m"))
+    est.train(_input_fn, max_steps=5)
+    exporter = Exporter(name="best")
+    export_path = os.path.join(str(tmp_path / "m"), exporter.name)
+    ckpt = os.path.join(str(tmp_path / "m"), "model.ckpt-5")
+    exporter.export(est, export_path, ckpt, {"loss": 0.1})
+    assert os.path.exists(os.path.join(export_path, "exported-5.pt"))
+
+
+def test_eval_monitor_hook_publishes_stats(tmp_path):
+    from tf_yarn_amd.estimator.metrics import EvalMonitorHook
+    from tf_yarn_amd.kv import KVClient, KVServer
+    server = KVServer()
+    try:
+        client = KVClient(server.address)
+        est = _make_estimator(str(tmp_path / "m"))
+        est.train(_input_fn, max_steps=3)
+        hook = EvalMonitorHook(client)
+        hook._task = "evaluator:0"
+        est.evaluate(_input_fn, steps=2, hooks=[hook])
+        stats = client.list("evaluator:0/")
+        assert "evaluator:0/nb_eval_steps" in stats
+        assert stats["evaluator:0/nb_eval_steps"] == b"2"
+        assert "evaluator:0/last_training_step" in stats
+        assert stats["evaluator:0/last_training_step"] == b"3"
+    finally:
+        server.stop()
+
+
+def test_step_per_second_hook_counts(tmp_path):
+    from tf_yarn_amd.estimator.metrics import StepPerSecondHook
+    est = _make_estimator(str(tmp_path / "m"))
+    hook = StepPerSecondHook(every_n_steps=2)
+    est.train(_input_fn, max_steps=6, hooks=[hook])
+    assert est.global_step == 6

@@ -142,3 +142,53 @@ def test_sharded_local_world1_matches_replicated():
     # backward + apply runs
     (out_buf.pow(2).mean() + wide_out.pow(2).mean()).backward()
     emb.apply_sparse_updates(0.1)
+
+
+def _worker4(rank, kv_addr, out_q):
+    from tf_yarn_amd.models.sharded_embedding import \
+        ShardedCriteoEmbeddings
+    from tf_yarn_amd.parallel import comm
+    client = KVClient(kv_addr)
+    comm.init_process_group(rank=rank, world_size=4, backend="gloo",
+                            kv_client=client)
+    try:
+        import torch as t
+        emb = ShardedCriteoEmbeddings([20] * 6, 4)  # 6 feats / 4 ranks
+        g = t.Generator().manual_seed(11)
+        full = t.randn(120, 4, generator=g)
+        own = t.cat([full[f * 20:(f + 1) * 20]
+                     for f in range(6) if f % 4 == rank])
+        emb.weight.data.copy_(own)
+        ids = t.randint(0, 20, (3, 6),
+                        generator=t.Generator().manual_seed(rank))
+        buf = t.zeros(3, 24)
+        out_buf, wide = emb(ids, buf, 0)
+        # reference: local gather on the full table (perm feature order)
+        offs = t.arange(6) * 20
+        perm = [f for s in range(4) for f in range(6) if f % 4 == s]
+        flat = (ids + offs)[:, perm].reshape(-1)
+        ref = full.index_select(0, flat).reshape(3, 24)
+        out_q.put((rank, bool(t.allclose(out_buf, ref, atol=1e-5))))
+    finally:
+        comm.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_sharded_forward_world4_uneven_features():
+    """6 features over 4 ranks (2,2,1,1): uneven alltoall splits."""
+    server = KVServer()
+    ctx = mp.get_context("spawn")
+    out_q = ctx.Queue()
+    procs = [ctx.Process(target=_worker4, args=(r, server.address, out_q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    try:
+        results = dict(out_q.get(timeout=150) for _ in range(4))
+    finally:
+        for p in procs:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.terminate()
+        server.stop()
+    assert all(results.values()), results
