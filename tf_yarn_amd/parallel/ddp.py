@@ -97,8 +97,14 @@ class BucketedDataParallel(nn.Module):
         self._hook_handles = []
         self._buckets: List[_Bucket] = []
         self._param_to_bucket: Dict[nn.Parameter, Tuple[_Bucket, int]] = {}
+        # Collectives must launch in the SAME order on every rank (NCCL
+        # contract).  Buckets therefore launch strictly in index order:
+        # a bucket whose grads complete early waits for its predecessors.
+        self._ready: List[bool] = []
+        self._next_launch = 0
         if self._world_size > 1:
             self._build_buckets()
+            self._ready = [False] * len(self._buckets)
             self._register_hooks()
             self._broadcast_module_states()
 
@@ -188,7 +194,14 @@ class BucketedDataParallel(nn.Module):
             param.grad = bucket.views[slot]
         bucket.pending -= 1
         if bucket.pending == 0:
-            self._launch_bucket(bucket)
+            self._ready[bucket.index] = True
+            self._launch_ready_in_order()
+
+    def _launch_ready_in_order(self) -> None:
+        while (self._next_launch < len(self._buckets)
+               and self._ready[self._next_launch]):
+            self._launch_bucket(self._buckets[self._next_launch])
+            self._next_launch += 1
 
     def _launch_bucket(self, bucket: _Bucket) -> None:
         op = dist.ReduceOp.AVG if (
@@ -207,21 +220,19 @@ class BucketedDataParallel(nn.Module):
     def _finalize_backward(self) -> None:
         self._backward_queued = False
         for bucket in self._buckets:
-            if bucket.pending != 0:
-                if bucket.pending == len(bucket.params):
-                    # whole bucket unused this iteration
-                    if self.find_unused_parameters:
-                        self._launch_bucket(bucket)
-                    else:
-                        bucket.pending = len(bucket.params)
-                        raise RuntimeError(
-                            f"bucket {bucket.index} has no ready grads; "
-                            "pass find_unused_parameters=True if the model "
-                            "has parameters unused in forward")
-                else:
-                    # partially-ready bucket: unused params keep their
-                    # (zero or stale-accumulated) view contents
-                    self._launch_bucket(bucket)
+            if bucket.pending != 0 \
+                    and bucket.pending == len(bucket.params) \
+                    and not self.find_unused_parameters:
+                bucket.pending = len(bucket.params)
+                raise RuntimeError(
+                    f"bucket {bucket.index} has no ready grads; "
+                    "pass find_unused_parameters=True if the model "
+                    "has parameters unused in forward")
+        # launch every not-yet-launched bucket (unused / partial grads
+        # keep their zero or stale-view contents), still in index order
+        while self._next_launch < len(self._buckets):
+            self._launch_bucket(self._buckets[self._next_launch])
+            self._next_launch += 1
         scale_needed = (self.average and
                         dist.get_backend(self.process_group) != "nccl")
         for bucket in self._buckets:
@@ -231,6 +242,8 @@ class BucketedDataParallel(nn.Module):
             if scale_needed:
                 bucket.buffer.div_(self._world_size)
             bucket.pending = len(bucket.params)
+        self._ready = [False] * len(self._buckets)
+        self._next_launch = 0
 
     # -- public API ----------------------------------------------------------
 
