@@ -104,7 +104,9 @@ def main() -> None:
                         embedding_dim=EMBEDDING_DIM, hidden=HIDDEN,
                         compute_dtype=compute_dtype,
                         sharded=True).to(device)
-    ddp = BucketedDataParallel(model) if world_size > 1 else model
+    # no BN-style buffers in this model: skip the per-forward broadcast
+    ddp = BucketedDataParallel(model, broadcast_buffers=False) \
+        if world_size > 1 else model
     module = ddp.module if world_size > 1 else model
 
     lr = 0.02

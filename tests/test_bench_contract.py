@@ -57,3 +57,18 @@ def test_bench_under_torchrun_2_ranks():
         cwd=REPO, env=env, capture_output=True, text=True, timeout=280)
     assert out.returncode == 0, out.stderr[-2000:]
     _check_json_line(out.stdout, 2)
+
+
+@pytest.mark.timeout(300)
+def test_bench_under_torchrun_4_ranks_uneven_shards():
+    """dp4: 26 features over 4 ranks (7,7,6,6) exercises the uneven
+    all-to-all splits the round-end scale run hits."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", "29547", "bench.py", "--gpus", "4",
+         "--steps", "2", "--warmup", "1", "--batch", "32",
+         "--table-rows", "500", "--dtype", "fp32"],
+        cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert out.returncode == 0, out.stderr[-2000:]
+    _check_json_line(out.stdout, 4)

@@ -266,13 +266,19 @@ class ShardedCriteoEmbeddings(nn.Module):
         offs = torch.tensor(
             [0] + list(torch.cumsum(torch.tensor(own_sizes), 0)[:-1]),
             dtype=torch.int64)
+        # PER-RANK buffers (different size/content on each rank): they must
+        # never be broadcast by the reducer — flagged _miyarn_sharded,
+        # which BucketedDataParallel skips.
+        offs._miyarn_sharded = True
         self.register_buffer("own_offsets", offs, persistent=True)
         # full offsets for the W == 1 local path (perm == identity there)
         full = torch.tensor(
             [0] + list(torch.cumsum(torch.tensor(table_sizes), 0)[:-1]),
             dtype=torch.int64)
         self.register_buffer("full_offsets", full, persistent=True)
-        self.register_buffer("own_offsets_tiled", torch.empty(0),
+        tiled0 = torch.empty(0, dtype=torch.int64)
+        tiled0._miyarn_sharded = True
+        self.register_buffer("own_offsets_tiled", tiled0,
                              persistent=False)
         self._deep_sink: List[Tuple[torch.Tensor, torch.Tensor]] = []
         self._wide_sink: List[Tuple[torch.Tensor, torch.Tensor]] = []
@@ -281,8 +287,9 @@ class ShardedCriteoEmbeddings(nn.Module):
     def _ensure_tiled(self, numel: int, device) -> None:
         if self.own_offsets_tiled.numel() < numel:
             reps = (numel + self.f_own - 1) // self.f_own
-            self.own_offsets_tiled = self.own_offsets.to(device).repeat(
-                reps)[:numel]
+            tiled = self.own_offsets.to(device).repeat(reps)[:numel]
+            tiled._miyarn_sharded = True  # per-rank: never broadcast
+            self.own_offsets_tiled = tiled
 
     def forward(self, ids: torch.Tensor, deep_out_buf: torch.Tensor,
                 col_offset: int

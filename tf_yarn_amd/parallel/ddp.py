@@ -168,6 +168,8 @@ class BucketedDataParallel(nn.Module):
                     p.detach().copy_(v)
                 bucket.buffer.zero_()
             for buf in self.module.buffers():
+                if getattr(buf, "_miyarn_sharded", False):
+                    continue  # per-rank buffer (sharded-embedding offsets)
                 dist.broadcast(buf, src=0, group=self.process_group)
             for p in self.module.parameters():
                 if getattr(p, "_miyarn_sparse", False) \
@@ -249,7 +251,8 @@ class BucketedDataParallel(nn.Module):
 
     def forward(self, *args, **kwargs):
         if self._world_size > 1 and self.broadcast_buffers:
-            bufs = list(self.module.buffers())
+            bufs = [b for b in self.module.buffers()
+                    if not getattr(b, "_miyarn_sharded", False)]
             if bufs:
                 with torch.no_grad():
                     for buf in bufs:
