@@ -129,3 +129,16 @@ def test_run_on_yarn_retries(nb_retries, nb_failures, expect_raise,
 def test_run_on_yarn_validates_topology():
     with pytest.raises(ValueError):
         run_on_yarn(lambda: None, {"worker": TaskSpec()})
+
+
+def test_tensorflow_alias_package():
+    """tf-yarn users import tf_yarn.tensorflow.*; the alias keeps that
+    import shape working against the estimator flavor."""
+    from tf_yarn_amd.tensorflow import (Experiment, KerasExperiment,
+                                        run_on_yarn)
+    import tf_yarn_amd.tensorflow.client as alias_client
+    import tf_yarn_amd.estimator.client as real_client
+    assert alias_client is real_client
+    assert run_on_yarn is real_client.run_on_yarn
+    from tf_yarn_amd.estimator.experiment import Experiment as E2
+    assert Experiment is E2
