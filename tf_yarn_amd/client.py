@@ -288,7 +288,6 @@ def _execute_and_await_termination(
         n_try=n_try)
 
     statuses: Dict[str, str] = {}
-    side_tasks = {"evaluator", "tensorboard", "ps"}
     while True:
         running = False
         failed = False
@@ -307,16 +306,12 @@ def _execute_and_await_termination(
             break
         if not running:
             break
-        # Training done => side tasks may linger (ps never terminates,
-        # reference _independent_workers_task.py:38-40); once every
-        # chief/worker has exited, shut the rest down.
-        train_done = all(
-            tp.process.poll() is not None
-            for tp in cluster.processes if tp.task.type not in side_tasks)
-        if train_done and any(
-                tp.task.type == "ps" and tp.process.poll() is None
-                for tp in cluster.processes):
-            pass  # the stop-barrier in the task modules handles ps exit
+        # Side tasks may outlive training by design: the evaluator keeps
+        # scanning checkpoints (20-min idle cap), tensorboard lingers its
+        # termination timeout, and ps exits once every worker says
+        # goodbye + the stop-barrier releases it (reference
+        # _independent_workers_task.py:38-43, tf_task_common.py:102-107).
+        # The poll loop therefore waits for ALL processes.
         time.sleep(poll_period)
 
     if failed:
