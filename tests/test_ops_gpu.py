@@ -402,3 +402,52 @@ def test_wgrad_nt128_matches_reference(B, N, M):
     err = (out - ref).abs().max().item()
     scale = ref.abs().max().item()
     assert err < 0.02 * max(1.0, scale), f"max err {err} scale {scale}"
+
+
+@requires_gpu
+def test_estimator_dnn_classifier_gpu():
+    """Estimator flavor on GPU: DNNClassifier + fused Adagrad, full
+    train/evaluate/checkpoint cycle on cuda."""
+    import tempfile
+
+    import torch as t
+
+    from tf_yarn_amd.estimator import DNNClassifier, RunConfig
+    from tf_yarn_amd.ops.optim import FusedAdagrad
+
+    model_dir = tempfile.mkdtemp(prefix="miyarn_est_gpu_")
+    est = DNNClassifier(
+        [32, 16], n_features=8, n_classes=2,
+        optimizer_fn=lambda p: FusedAdagrad(p, lr=0.05),
+        model_dir=model_dir, config=RunConfig(save_checkpoints_steps=20),
+        device="cuda")
+
+    def input_fn():
+        t.manual_seed(0)
+        for _ in range(30):
+            x = t.randn(64, 8)
+            yield x, (x.sum(dim=1) > 0).long()
+
+    est.train(input_fn, max_steps=60)
+    result = est.evaluate(input_fn, steps=10)
+    assert result["accuracy"] > 0.8, result
+    assert est.latest_checkpoint().endswith("model.ckpt-60")
+
+
+@requires_gpu
+def test_keras_model_gpu_fused_adadelta():
+    """Keras shim on GPU with the fused Adadelta optimizer (the README's
+    Keras optimizer)."""
+    import torch as t
+    from torch import nn
+
+    from tf_yarn_amd.estimator.keras import KerasModel
+
+    t.manual_seed(0)
+    model = KerasModel(nn.Sequential(nn.Linear(8, 32), nn.ReLU(),
+                                     nn.Linear(32, 1))).to("cuda")
+    model.compile(optimizer="adadelta", loss="mse")
+    x = t.randn(256, 8)
+    y = x.sum(dim=1, keepdim=True)
+    hist = model.fit(x, y, epochs=5, batch_size=32)
+    assert hist["loss"][-1] < hist["loss"][0]
