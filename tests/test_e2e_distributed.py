@@ -1,0 +1,55 @@
+"""End-to-end tests for the raw `distributed` flavor and the env-check
+CLI round trip (reference check_hadoop_env submits a 1-container app)."""
+
+import sys
+
+import cloudpickle
+import pytest
+import torch
+
+from tf_yarn_amd import TaskSpec
+from tf_yarn_amd.distributed import TaskParameters, run_on_yarn
+
+cloudpickle.register_pickle_by_value(sys.modules[__name__])
+
+
+def _train_fn(task_params: TaskParameters):
+    import os
+
+    import torch
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = task_params.master_address
+    os.environ["MASTER_PORT"] = str(task_params.master_port)
+    dist.init_process_group("gloo", rank=task_params.rank,
+                            world_size=task_params.world_size)
+    t = torch.tensor([float(task_params.rank + 1)])
+    dist.all_reduce(t)
+    expected = sum(range(1, task_params.world_size + 1))
+    assert t.item() == expected, (t.item(), expected)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_distributed_flavor_allreduce(tmp_path):
+    metrics = run_on_yarn(
+        _train_fn,
+        {
+            "chief": TaskSpec(memory=512, vcores=1),
+            "worker": TaskSpec(memory=512, vcores=1, instances=2),
+        },
+        base_dir=str(tmp_path / "app"),
+    )
+    assert metrics is not None
+    assert metrics.total_training_duration is not None
+
+
+@pytest.mark.timeout(180)
+def test_check_env_round_trip(tmp_path):
+    from tf_yarn_amd.bin.check_env import check_local_env, \
+        launch_remote_check
+    local = check_local_env()
+    assert "torch" in local
+    assert local["gloo_backend"] is True
+    assert launch_remote_check(str(tmp_path / "app")) is True
