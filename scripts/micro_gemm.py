@@ -99,6 +99,18 @@ def custom_wgrad():
             gf = 2 * B * cin * cout / 1e9
             print(f"[wgrad128 {cout}x{cin} sk={sk:2}] {t:7.1f}us "
                   f"({gf/t*1e3:5.0f} TF)")
+    for (cin, cout) in [(432, 1024), (1024, 512), (512, 256)]:
+        x = torch.randn(B, cin, device="cuda").to(torch.bfloat16)
+        dy = torch.randn(B, cout, device="cuda").to(torch.bfloat16)
+        gf = 2 * B * cin * cout / 1e9
+        for sk in (8, 16, 32, 64):
+            try:
+                t = timeit(lambda: C.wgrad_nt256(dy, x, sk))
+                print(f"[wgrad256 {cout}x{cin} sk={sk:2}] {t:7.1f}us "
+                      f"({gf/t*1e3:5.0f} TF)")
+            except RuntimeError as e:
+                print(f"[wgrad256 {cout}x{cin}] skipped: {e}")
+                break
     x = torch.randn(B, 256, device="cuda").to(torch.bfloat16)
     dy = torch.randn(B, device="cuda").to(torch.bfloat16)
     t = timeit(lambda: C.col_reduce_dot(x, dy))

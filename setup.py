@@ -35,6 +35,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "elementwise.hip"),
         os.path.join(CSRC, "wgrad.hip"),
         os.path.join(CSRC, "wgrad128.hip"),
+        os.path.join(CSRC, "wgrad256.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3"],

@@ -451,3 +451,19 @@ def test_keras_model_gpu_fused_adadelta():
     y = x.sum(dim=1, keepdim=True)
     hist = model.fit(x, y, epochs=5, batch_size=32)
     assert hist["loss"][-1] < hist["loss"][0]
+
+
+@requires_gpu
+@pytest.mark.parametrize("B,N,M", [(4096, 256, 128), (8192, 1024, 432),
+                                   (65536, 256, 512), (4096, 256, 264)])
+def test_wgrad_nt256_matches_reference(B, N, M):
+    import tf_yarn_amd.ops._C as C
+    torch.manual_seed(18)
+    dy = (torch.randn(B, N, device="cuda") / 8).to(torch.bfloat16)
+    x = (torch.randn(B, M, device="cuda") / 8).to(torch.bfloat16)
+    out = C.wgrad_nt256(dy, x, 0)
+    ref = dy.float().t().mm(x.float())
+    assert out.shape == (N, M)
+    err = (out - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err < 0.02 * max(1.0, scale), f"max err {err} scale {scale}"

@@ -54,6 +54,7 @@ torch::Tensor col_reduce_dot(torch::Tensor x, torch::Tensor dy);
 // wgrad.hip
 torch::Tensor wgrad_nt(torch::Tensor dy, torch::Tensor x, int64_t splitk);
 torch::Tensor wgrad_nt128(torch::Tensor dy, torch::Tensor x, int64_t splitk);
+torch::Tensor wgrad_nt256(torch::Tensor dy, torch::Tensor x, int64_t splitk);
 void convert_scaled(torch::Tensor src, torch::Tensor dst, double scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -75,6 +76,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Split-K MFMA weight gradient: dW = dy^T @ x (bf16 in, fp32 out)");
   m.def("wgrad_nt128", &wgrad_nt128,
         "Split-K MFMA weight gradient, 128x128 tiles + XOR-swizzled LDS");
+  m.def("wgrad_nt256", &wgrad_nt256,
+        "Split-K MFMA weight gradient, 256x256 tiles (8 waves)");
   m.def("col_reduce_dot", &col_reduce_dot,
         "dw[m] = sum_b dy[b] * x[b,m] (single-logit head wgrad)");
   m.def("bias_relu_bwd_db", &bias_relu_bwd_db,
