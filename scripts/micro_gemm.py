@@ -103,7 +103,7 @@ def custom_wgrad():
         x = torch.randn(B, cin, device="cuda").to(torch.bfloat16)
         dy = torch.randn(B, cout, device="cuda").to(torch.bfloat16)
         gf = 2 * B * cin * cout / 1e9
-        for sk in (8, 16, 32, 64):
+        for sk in (24, 28, 32, 36, 40, 48):
             try:
                 t = timeit(lambda: C.wgrad_nt256(dy, x, sk))
                 print(f"[wgrad256 {cout}x{cin} sk={sk:2}] {t:7.1f}us "

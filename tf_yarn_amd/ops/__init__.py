@@ -311,16 +311,24 @@ def bias_relu(x: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
     return BiasReLU.apply(x, bias)
 
 
-def _use_custom_wgrad(dz: torch.Tensor, x: torch.Tensor) -> bool:
-    """Route the weight gradient through the split-K MFMA kernel where it
-    measured faster than hipBLASLt (small outputs: N*M <= 256*512; the
-    library's larger-tile solutions win on the bigger layers)."""
+def _custom_wgrad_kernel(dz: torch.Tensor, x: torch.Tensor):
+    """Pick the split-K MFMA wgrad variant for this shape, or None to use
+    hipBLASLt.  Measured (micro_gemm.py, B=65536): 256^2 tiles reach
+    312/359 TF on the 1024x432 / 512x1024 layers (hipBLASLt in-context ran
+    them at ~207 us each); 128^2 tiles win the small 256x512 layer
+    (94 us vs ~142-190)."""
     if not (dz.is_cuda and HAVE_EXT):
-        return False
+        return None
+    if dz.dtype != torch.bfloat16 or x.dtype != torch.bfloat16:
+        return None
     n, m = dz.size(1), x.size(1)
-    return (dz.dtype == torch.bfloat16 and x.dtype == torch.bfloat16
-            and n % 128 == 0 and m % 8 == 0 and dz.size(0) % 64 == 0
-            and n * m <= 256 * 512)
+    if m % 8 != 0 or dz.size(0) % 64 != 0:
+        return None
+    if n % 256 == 0 and n * m > 256 * 512:
+        return _C.wgrad_nt256
+    if n % 128 == 0 and n * m <= 256 * 512:
+        return _C.wgrad_nt128
+    return None
 
 
 class LinearBiasReLU(torch.autograd.Function):
@@ -347,8 +355,9 @@ class LinearBiasReLU(torch.autograd.Function):
             dz = bias_relu_bwd(dy, y)
             dbias = dz.sum(dim=tuple(range(dz.dim() - 1)))
         dx = dz.matmul(weight)
-        if _use_custom_wgrad(dz, x):
-            dw = _C.wgrad_nt128(dz, x, 0).to(weight.dtype)
+        kernel = _custom_wgrad_kernel(dz, x)
+        if kernel is not None:
+            dw = kernel(dz, x, 0).to(weight.dtype)
         else:
             dw = dz.t().matmul(x)
         return dx, dw, dbias
