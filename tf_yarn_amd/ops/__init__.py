@@ -317,6 +317,9 @@ def _custom_wgrad_kernel(dz: torch.Tensor, x: torch.Tensor):
     312/359 TF on the 1024x432 / 512x1024 layers (hipBLASLt in-context ran
     them at ~207 us each); 128^2 tiles win the small 256x512 layer
     (94 us vs ~142-190)."""
+    import os
+    if os.environ.get("MIYARN_WGRAD") == "lib":
+        return None  # A/B escape hatch: force hipBLASLt
     if not (dz.is_cuda and HAVE_EXT):
         return None
     if dz.dtype != torch.bfloat16 or x.dtype != torch.bfloat16:
