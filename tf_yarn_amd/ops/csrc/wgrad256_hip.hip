@@ -31,7 +31,10 @@ union U16x4c {
 
 __device__ __forceinline__ int lds_off4(int n, int k) {
   int byte = n * 128 + k * 2;
-  return byte ^ ((n & 7) << 4);
+  // combined swizzle: (n&7) spreads the b128 READ groups (consecutive
+  // rows), (n>>3)&7 spreads the b64 WRITE groups (lanes stride 8 rows —
+  // with the plain (n&7) term those writes were a 16-way conflict)
+  return byte ^ (((n ^ (n >> 3)) & 7) << 4);
 }
 
 __global__ __launch_bounds__(512)
