@@ -186,7 +186,10 @@ torch::Tensor wgrad_nt128(torch::Tensor dy, torch::Tensor x,
               "wgrad_nt128 needs N % 128 == 0, M % 8 == 0, B % 64 == 0");
   const int tiles = (N / W3_BN) * ((M + W3_BM - 1) / W3_BM);
   if (splitk <= 0)
-    splitk = std::max<int64_t>(1, 512 / std::max(1, tiles));
+    // sk sweep (scripts/micro_gemm.py): >32 splits exceed one residency
+    // wave of blocks and cliff; 32 is the measured optimum
+    splitk = std::min<int64_t>(
+        32, std::max<int64_t>(1, 512 / std::max(1, tiles)));
   int64_t chunk = ((B + splitk - 1) / splitk + W3_BK - 1) / W3_BK * W3_BK;
   splitk = (B + chunk - 1) / chunk;
   auto part = torch::empty({splitk, N, M},

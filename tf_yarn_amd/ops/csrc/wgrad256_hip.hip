@@ -172,7 +172,9 @@ torch::Tensor wgrad_nt256(torch::Tensor dy, torch::Tensor x,
               "wgrad_nt256 needs N % 256 == 0, M % 8 == 0, B % 64 == 0");
   const int tiles = (N / W4_BN) * ((M + W4_BM - 1) / W4_BM);
   if (splitk <= 0)
-    splitk = std::max<int64_t>(1, 256 / std::max(1, tiles));
+    // 8-wave blocks: 256 blocks = one full residency wave; sk > 32 cliffs
+    splitk = std::min<int64_t>(
+        32, std::max<int64_t>(1, 256 / std::max(1, tiles)));
   int64_t chunk = ((B + splitk - 1) / splitk + W4_BK - 1) / W4_BK * W4_BK;
   splitk = (B + chunk - 1) / chunk;
   auto part = torch::empty({splitk, N, M},
