@@ -467,3 +467,33 @@ def test_wgrad_nt256_matches_reference(B, N, M):
     err = (out - ref).abs().max().item()
     scale = ref.abs().max().item()
     assert err < 0.02 * max(1.0, scale), f"max err {err} scale {scale}"
+
+
+@requires_gpu
+def test_emb_bwd_sgd_sorted_heavy_collisions():
+    """The atomic-free sorted path must exactly sum duplicate-id grads
+    (run-head segmented reduction, no atomics)."""
+    import tf_yarn_amd.ops._C as C
+    torch.manual_seed(19)
+    table = torch.randn(50, 16).cuda()  # 50 rows, 65536 updates: runs ~1300
+    ref = table.clone()
+    ids = torch.randint(0, 50, (65536,)).cuda()
+    grad = torch.randn(65536, 16).cuda()
+    sorted_ids, perm = torch.sort(ids)
+    g_sorted = grad.index_select(0, perm).contiguous()
+    C.emb_bwd_sgd_sorted(table, sorted_ids, g_sorted, 0.1, 0.5)
+    ref.index_add_(0, ids, grad, alpha=-0.05)
+    assert torch.allclose(table, ref, atol=1e-2, rtol=1e-3)
+
+
+@requires_gpu
+def test_emb_bwd_sgd_wrapper_sorted_path_matches_atomic():
+    import tf_yarn_amd.ops._C as C
+    torch.manual_seed(20)
+    table_a = torch.randn(1000, 16).cuda()
+    table_b = table_a.clone()
+    ids = torch.randint(0, 1000, (8192,)).cuda()
+    grad = torch.randn(8192, 16).cuda().to(torch.bfloat16)
+    ops.emb_bwd_sgd(table_a, ids, grad, lr=0.2, scale=1.0)  # sorted path
+    C.emb_bwd_sgd(table_b, ids, grad, 0.2, 1.0)             # atomic path
+    assert torch.allclose(table_a, table_b, atol=1e-3)

@@ -34,6 +34,8 @@ void fused_adadelta(torch::Tensor param, torch::Tensor grad,
 torch::Tensor emb_fwd(torch::Tensor table, torch::Tensor ids, bool out_bf16);
 void emb_bwd_sgd(torch::Tensor table, torch::Tensor ids, torch::Tensor grad,
                  double lr, double scale);
+void emb_bwd_sgd_sorted(torch::Tensor table, torch::Tensor sorted_ids,
+                        torch::Tensor grad, double lr, double scale);
 void emb_bwd_dense(torch::Tensor grad_table, torch::Tensor ids,
                    torch::Tensor grad, double scale);
 
@@ -68,6 +70,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("emb_fwd", &emb_fwd, "Fused multi-table embedding gather");
   m.def("emb_bwd_sgd", &emb_bwd_sgd,
         "Fused sparse embedding grad scatter + SGD update");
+  m.def("emb_bwd_sgd_sorted", &emb_bwd_sgd_sorted,
+        "Atomic-free segmented scatter+SGD over SORTED ids");
   m.def("emb_bwd_dense", &emb_bwd_dense,
         "Sparse embedding grad scatter into dense grad table");
   m.def("bias_relu_fwd", &bias_relu_fwd, "Fused bias+ReLU forward");

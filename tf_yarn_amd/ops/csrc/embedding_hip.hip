@@ -182,6 +182,40 @@ __global__ void emb_fwd_into_kernel(const float* __restrict__ table,
   }
 }
 
+// Sorted segmented scatter+SGD: ids are SORTED so duplicates are
+// contiguous; only the head thread of each run applies the (summed)
+// update with a plain read-modify-write — no atomics (measured 4.6x
+// faster than atomicAdd at the bench shape; see scripts/micro_scatter).
+template <typename GIo>
+__global__ void emb_bwd_sgd_sorted_kernel(
+    float* __restrict__ table, const int64_t* __restrict__ ids,
+    const typename GIo::scalar_t* __restrict__ g,
+    int64_t n_rows, int64_t dim, float neg_lr_scale) {
+  const int64_t dvec = dim >> 2;
+  const int64_t total = n_rows * dvec;
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       t < total; t += stride) {
+    const int64_t row = t / dvec;
+    const int64_t c4 = t - row * dvec;
+    const int64_t id = ids[row];
+    if (row > 0 && ids[row - 1] == id) continue;  // not a run head
+    float acc[4];
+    QuadIo<GIo>::load4(g, row * dvec + c4, acc);
+    for (int64_t r = row + 1; r < n_rows && ids[r] == id; ++r) {
+      float more[4];
+      QuadIo<GIo>::load4(g, r * dvec + c4, more);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[j] += more[j];
+    }
+    float* dst = table + id * dim + c4 * 4;
+    f32x4 cur = *reinterpret_cast<f32x4*>(dst);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) cur[j] += neg_lr_scale * acc[j];
+    *reinterpret_cast<f32x4*>(dst) = cur;
+  }
+}
+
 void check_emb(const torch::Tensor& table, const torch::Tensor& ids,
                int64_t dim) {
   TORCH_CHECK(table.is_cuda() && table.is_contiguous() &&
@@ -401,5 +435,35 @@ void emb_fwd_into(torch::Tensor table, torch::Tensor ids,
                        table.data_ptr<float>(), ids.data_ptr<int64_t>(),
                        out.data_ptr<float>(), n, dim, F,
                        out.size(1) / 4, col_offset / 4);
+  }
+}
+
+
+void emb_bwd_sgd_sorted(torch::Tensor table, torch::Tensor sorted_ids,
+                        torch::Tensor grad, double lr, double scale) {
+  const int64_t dim = table.size(1);
+  const int64_t n = sorted_ids.numel();
+  check_emb(table, sorted_ids, dim);
+  TORCH_CHECK(dim % 4 == 0, "sorted scatter needs dim % 4 == 0");
+  TORCH_CHECK(grad.is_cuda() && grad.is_contiguous() &&
+              grad.numel() == n * dim, "grad shape mismatch");
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int grid = miyarn_grid_cap(n * (dim / 4), 8192);
+  float nls = (float)(-lr * scale);
+  if (grad.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(emb_bwd_sgd_sorted_kernel<F32Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream,
+                       table.data_ptr<float>(),
+                       sorted_ids.data_ptr<int64_t>(),
+                       grad.data_ptr<float>(), n, dim, nls);
+  } else {
+    TORCH_CHECK(grad.scalar_type() == torch::kBFloat16,
+                "grad must be fp32 or bf16");
+    hipLaunchKernelGGL(emb_bwd_sgd_sorted_kernel<Bf16Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream,
+                       table.data_ptr<float>(),
+                       sorted_ids.data_ptr<int64_t>(),
+                       reinterpret_cast<unsigned short*>(grad.data_ptr()),
+                       n, dim, nls);
   }
 }
