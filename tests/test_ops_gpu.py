@@ -494,6 +494,6 @@ def test_emb_bwd_sgd_wrapper_sorted_path_matches_atomic():
     table_b = table_a.clone()
     ids = torch.randint(0, 1000, (8192,)).cuda()
     grad = torch.randn(8192, 16).cuda().to(torch.bfloat16)
-    ops.emb_bwd_sgd(table_a, ids, grad, lr=0.2, scale=1.0)  # sorted path
+    ops.emb_bwd_sgd(table_a, ids, grad, lr=0.2, scale=1.0)  # wrapper
     C.emb_bwd_sgd(table_b, ids, grad, 0.2, 1.0)             # atomic path
     assert torch.allclose(table_a, table_b, atol=1e-3)

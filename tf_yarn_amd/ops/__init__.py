@@ -165,22 +165,15 @@ def emb_bwd_sgd(table: torch.Tensor, ids: torch.Tensor, grad: torch.Tensor,
                 *, lr: float, scale: float = 1.0) -> None:
     """table[ids[i], :] -= lr * scale * grad[i, :] (fused sparse update).
 
-    On GPU for vectorizable dims the ids are sorted first and applied with
-    the atomic-free segmented kernel (plain RMW on run heads — atomics
-    measured 4.6x slower at the bench shape); small/odd-dim cases use the
-    atomic scatter."""
+    The atomic scatter is the default: the atomic-free sorted kernel
+    (`_C.emb_bwd_sgd_sorted`) is 4.6x faster per-kernel, but paying
+    torch.sort + a 54 MB gather every step measured NET-slower at the
+    bench shape (bench 2.30 vs 1.86 ms/step) — use it only when ids
+    arrive pre-sorted."""
     if _on_gpu(table, ids, grad):
         _require_ext()
-        flat_ids = ids.reshape(-1)
-        dim = table.size(1)
-        if dim % 4 == 0 and flat_ids.numel() >= 4096:
-            sorted_ids, perm = torch.sort(flat_ids)
-            g_sorted = grad.reshape(flat_ids.numel(), dim) \
-                .index_select(0, perm).contiguous()
-            _C.emb_bwd_sgd_sorted(table, sorted_ids, g_sorted, lr, scale)
-            return
-        _C.emb_bwd_sgd(table, flat_ids.contiguous(), grad.contiguous(),
-                       lr, scale)
+        _C.emb_bwd_sgd(table, ids.reshape(-1).contiguous(),
+                       grad.contiguous(), lr, scale)
         return
     table.index_add_(0, ids.reshape(-1),
                      grad.reshape(ids.numel(), -1).float(),
