@@ -36,6 +36,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "wgrad.hip"),
         os.path.join(CSRC, "wgrad128.hip"),
         os.path.join(CSRC, "wgrad256.hip"),
+        os.path.join(CSRC, "gemm_bt.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3"],

@@ -59,6 +59,10 @@ torch::Tensor wgrad_nt128(torch::Tensor dy, torch::Tensor x, int64_t splitk);
 torch::Tensor wgrad_nt256(torch::Tensor dy, torch::Tensor x, int64_t splitk);
 void convert_scaled(torch::Tensor src, torch::Tensor dst, double scale);
 
+// gemm_bt.hip
+torch::Tensor gemm_bt(torch::Tensor a, torch::Tensor b,
+                      c10::optional<torch::Tensor> bias, bool relu);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "tf_yarn_amd MI355X (gfx950) HIP kernels";
   m.def("fused_sgd", &fused_sgd, "Fused SGD(+momentum) step");
@@ -93,4 +97,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("emb_scatter_sum", &emb_scatter_sum,
         "Wide-part scatter: table[ids[b,f]] += alpha * g[b]");
   m.def("convert_scaled", &convert_scaled, "Scaled bf16<->fp32 convert");
+  m.def("gemm_bt", &gemm_bt,
+        "C = A @ B^T (both K-major bf16) with fused bias+ReLU epilogue");
 }
