@@ -142,3 +142,15 @@ def test_linear_bias_relu_matches_autograd():
     assert torch.allclose(x.grad, x2.grad, atol=1e-5)
     assert torch.allclose(w.grad, w2.grad, atol=1e-5)
     assert torch.allclose(b.grad, b2.grad, atol=1e-5)
+
+
+def test_gemm_bt_cpu_fallback():
+    torch.manual_seed(23)
+    a = torch.randn(32, 24)
+    b = torch.randn(16, 24)
+    bias = torch.randn(16)
+    out = ops.gemm_bt(a, b, bias, relu=True)
+    ref = torch.relu(a @ b.t() + bias)
+    assert torch.allclose(out, ref, atol=1e-5)
+    out2 = ops.gemm_bt(a, b, None, relu=False)
+    assert torch.allclose(out2, a @ b.t(), atol=1e-5)

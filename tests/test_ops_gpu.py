@@ -497,3 +497,56 @@ def test_emb_bwd_sgd_wrapper_sorted_path_matches_atomic():
     ops.emb_bwd_sgd(table_a, ids, grad, lr=0.2, scale=1.0)  # wrapper
     C.emb_bwd_sgd(table_b, ids, grad, 0.2, 1.0)             # atomic path
     assert torch.allclose(table_a, table_b, atol=1e-3)
+
+
+@requires_gpu
+@pytest.mark.parametrize("M,N,K,relu", [
+    (4096, 1024, 432, True),    # L1 forward (ragged K tail: 432 % 64 != 0)
+    (4096, 512, 1024, True),    # L2 forward
+    (4096, 256, 512, False),    # L3 shape, no epilogue
+    (4096, 432, 1024, False),   # L1 dgrad (ragged N: B rows masked)
+])
+def test_gemm_bt_matches_reference(M, N, K, relu):
+    import tf_yarn_amd.ops._C as C
+    torch.manual_seed(21)
+    a = (torch.randn(M, K, device="cuda") / 8).to(torch.bfloat16)
+    b = (torch.randn(N, K, device="cuda") / 8).to(torch.bfloat16)
+    bias = torch.randn(N, device="cuda").to(torch.bfloat16) if relu else None
+    out = C.gemm_bt(a, b, bias, relu)
+    ref = a.float() @ b.float().t()
+    if bias is not None:
+        ref = ref + bias.float()
+    if relu:
+        ref = torch.relu(ref)
+    assert out.shape == (M, N) and out.dtype == torch.bfloat16
+    err = (out.float() - ref).abs().max().item()
+    scale = max(1.0, ref.abs().max().item())
+    assert err < 0.03 * scale, f"max err {err} scale {scale}"
+
+
+@requires_gpu
+def test_linear_bias_relu_custom_fwd_gpu():
+    """The fused gemm_bt forward path must agree with the lib path."""
+    import os
+    torch.manual_seed(22)
+    x = (torch.randn(512, 432, device="cuda") / 4).to(
+        torch.bfloat16).requires_grad_(True)
+    w = (torch.randn(1024, 432, device="cuda") / 8).to(
+        torch.bfloat16).requires_grad_(True)
+    b = torch.randn(1024, device="cuda").to(torch.bfloat16).requires_grad_(True)
+    os.environ["MIYARN_FWD"] = "custom"  # fused path is opt-in (slower)
+    try:
+        y = ops.linear_bias_relu(x, w, b)
+        y.float().pow(2).sum().backward()
+    finally:
+        del os.environ["MIYARN_FWD"]
+    x2 = x.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    y2 = ops.linear_bias_relu(x2, w2, b2)
+    y2.float().pow(2).sum().backward()
+    assert torch.allclose(y.float(), y2.float(), atol=0.05, rtol=0.05)
+    assert torch.allclose(x.grad.float(), x2.grad.float(),
+                          atol=0.1, rtol=0.1)
+    assert torch.allclose(w.grad.float(), w2.grad.float(),
+                          atol=0.5, rtol=0.1)

@@ -267,6 +267,22 @@ def col_reduce_dot(x: torch.Tensor, dy: torch.Tensor) -> torch.Tensor:
     return (x.float() * dy.float().unsqueeze(1)).sum(dim=0)
 
 
+def gemm_bt(a: torch.Tensor, b: torch.Tensor,
+            bias: "torch.Tensor | None" = None,
+            relu: bool = False) -> torch.Tensor:
+    """C[M, N] = A[M, K] @ B[N, K]^T (nn.Linear forward form), optional
+    fused bias+ReLU epilogue. Both operands K-major bf16 on GPU."""
+    if _on_gpu(a, b):
+        _require_ext()
+        return _C.gemm_bt(a.contiguous(), b.contiguous(), bias, relu)
+    out = a.float() @ b.float().t()
+    if bias is not None:
+        out = out + bias.float()
+    if relu:
+        out = torch.relu(out)
+    return out.to(a.dtype)
+
+
 class ScalarHeadFn(torch.autograd.Function):
     """y[b] = x[b].w + bias for a single-logit head; the wgrad runs as a
     streaming column reduction instead of hipBLASLt's M=1 GEMM
@@ -341,6 +357,29 @@ def _custom_wgrad_kernel(dz: torch.Tensor, x: torch.Tensor):
     return None
 
 
+def _custom_fwd_ok(x: torch.Tensor, weight: torch.Tensor,
+                   bias: torch.Tensor) -> bool:
+    """True when gemm_bt should fuse the whole forward (GEMM + bias +
+    ReLU in one kernel).  OPT-IN (MIYARN_FWD=custom): measured on MI355X
+    the fused kernel runs at 361-613 TF on the MLP forward shapes while
+    hipBLASLt + the separate bias_relu kernel reach 444-831 TF combined —
+    with only 7-16 K-tile iterations the software pipeline never hits the
+    steady state the wgrad kernels reach over B=65536 (scripts/
+    micro_gemm.py --fwd)."""
+    import os
+    if os.environ.get("MIYARN_FWD") != "custom":
+        return False
+    if not (x.is_cuda and HAVE_EXT):
+        return False
+    if x.dtype != torch.bfloat16 or weight.dtype != torch.bfloat16 \
+            or bias.dtype != torch.bfloat16:
+        return False
+    if x.dim() != 2 or not x.is_contiguous() or not weight.is_contiguous():
+        return False
+    m, k = x.shape
+    return m % 256 == 0 and k % 16 == 0 and weight.size(0) % 16 == 0
+
+
 class LinearBiasReLU(torch.autograd.Function):
     """y = relu(x @ w.T + bias) with a fully-controlled backward:
     fused dx+dbias kernel, dgrad via hipBLASLt, wgrad via the custom
@@ -348,8 +387,11 @@ class LinearBiasReLU(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias):
-        z = x.matmul(weight.t())
-        y = bias_relu_fwd(z, bias)
+        if _custom_fwd_ok(x, weight, bias):
+            y = _C.gemm_bt(x, weight, bias, True)
+        else:
+            z = x.matmul(weight.t())
+            y = bias_relu_fwd(z, bias)
         ctx.save_for_backward(x, weight, y)
         return y
 
