@@ -550,3 +550,24 @@ def test_linear_bias_relu_custom_fwd_gpu():
                           atol=0.1, rtol=0.1)
     assert torch.allclose(w.grad.float(), w2.grad.float(),
                           atol=0.5, rtol=0.1)
+
+
+@requires_gpu
+@pytest.mark.parametrize("rows,cols,dtype", [
+    (4096, 256, torch.bfloat16), (1000, 64, torch.float32),
+    (512, 512, torch.bfloat16), (333, 12, torch.bfloat16),
+])
+def test_row_dot_matches_reference(rows, cols, dtype):
+    import tf_yarn_amd.ops._C as C
+    torch.manual_seed(24)
+    x = (torch.randn(rows, cols, device="cuda") / 4).to(dtype)
+    w = (torch.randn(cols, device="cuda") / 4).to(dtype)
+    bias = torch.randn(1, device="cuda").to(dtype)
+    ref = x.float() @ w.float() + bias.float()
+    out = ops.row_dot(x, w, bias)
+    assert out.shape == (rows,) and out.dtype == dtype
+    err = (out.float() - ref).abs().max().item()
+    assert err < 0.02 * max(1.0, ref.abs().max().item())
+    out2 = ops.row_dot(x, w, None)
+    assert torch.allclose(out2.float(), (x.float() @ w.float()),
+                          atol=0.02 * max(1.0, ref.abs().max().item()))

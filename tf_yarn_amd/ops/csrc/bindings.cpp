@@ -52,6 +52,8 @@ torch::Tensor bias_relu_bwd(torch::Tensor dy, torch::Tensor y);
 std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
                                             torch::Tensor y);
 torch::Tensor col_reduce_dot(torch::Tensor x, torch::Tensor dy);
+torch::Tensor row_dot(torch::Tensor x, torch::Tensor w,
+                      c10::optional<torch::Tensor> bias);
 
 // wgrad.hip
 torch::Tensor wgrad_nt(torch::Tensor dy, torch::Tensor x, int64_t splitk);
@@ -88,6 +90,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Split-K MFMA weight gradient, 256x256 tiles (8 waves)");
   m.def("col_reduce_dot", &col_reduce_dot,
         "dw[m] = sum_b dy[b] * x[b,m] (single-logit head wgrad)");
+  m.def("row_dot", &row_dot,
+        "y[b] = x[b].w + bias (single-logit head forward GEMV)");
   m.def("bias_relu_bwd_db", &bias_relu_bwd_db,
         "Fused ReLU backward + dbias reduction (returns [dx, dbias_fp32])");
   m.def("emb_fwd_into", &emb_fwd_into,
