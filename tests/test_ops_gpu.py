@@ -571,3 +571,22 @@ def test_row_dot_matches_reference(rows, cols, dtype):
     out2 = ops.row_dot(x, w, None)
     assert torch.allclose(out2.float(), (x.float() @ w.float()),
                           atol=0.02 * max(1.0, ref.abs().max().item()))
+
+
+@requires_gpu
+def test_scalar_head_ragged_width_padded_path():
+    """ScalarHead(13) on GPU (padded row_dot path) must match the plain
+    matmul reference, including grads."""
+    from tf_yarn_amd.models.wide_deep import ScalarHead
+    torch.manual_seed(25)
+    head = ScalarHead(13, dtype=torch.bfloat16).cuda()
+    x = (torch.randn(512, 13, device="cuda") / 4).to(torch.bfloat16)
+    y = head(x)
+    ref = x @ head.weight + head.bias
+    assert torch.allclose(y.float(), ref.float(), atol=0.02)
+    y.float().sum().backward()
+    gw = head.weight.grad.clone()
+    head.zero_grad()
+    (x @ head.weight + head.bias).float().sum().backward()
+    assert torch.allclose(gw.float(), head.weight.grad.float(),
+                          atol=0.05, rtol=0.05)

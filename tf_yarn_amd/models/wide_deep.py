@@ -296,6 +296,15 @@ class ScalarHead(nn.Module):
     def forward(self, x):
         if x.size(-1) % 4 == 0:
             return ops.ScalarHeadFn.apply(x, self.weight, self.bias)
+        if x.is_cuda:
+            # Ragged width (e.g. the 13 raw dense features): hipBLASLt
+            # runs this bias-GEMV at ~9 GB/s (192 us measured at
+            # 65536x13); zero-pad to a quad boundary and take the
+            # streaming row_dot path instead (~2 us pad + ~5 us dot).
+            pad = (-x.size(-1)) % 4
+            xp = nn.functional.pad(x, (0, pad))
+            wp = nn.functional.pad(self.weight, (0, pad))
+            return ops.ScalarHeadFn.apply(xp, wp, self.bias)
         return x @ self.weight + self.bias
 
 

@@ -304,6 +304,7 @@ class ScalarHeadFn(torch.autograd.Function):
         ctx.save_for_backward(x, weight)
         if _on_gpu(x, weight) and x.size(1) % 4 == 0 \
                 and x.size(1) <= 512 and bias.numel() == 1:
+            _require_ext()
             return _C.row_dot(x.contiguous(), weight.contiguous(),
                               bias.reshape(1))
         return x @ weight + bias
@@ -312,9 +313,13 @@ class ScalarHeadFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, weight = ctx.saved_tensors
         dy = dy.contiguous()
-        dw = col_reduce_dot(x, dy).to(weight.dtype)
-        dx = dy.unsqueeze(1) * weight.unsqueeze(0)
-        db = dy.sum().reshape(1).to(weight.dtype)
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:  # input data (wide head) needs no dx
+            dx = dy.unsqueeze(1) * weight.unsqueeze(0)
+        if ctx.needs_input_grad[1]:
+            dw = col_reduce_dot(x, dy).to(weight.dtype)
+        if ctx.needs_input_grad[2]:
+            db = dy.sum().reshape(1).to(weight.dtype)
         return dx, dw, db
 
 
