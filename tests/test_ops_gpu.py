@@ -228,9 +228,11 @@ def test_fused_sgd_mt_many_tensors_chunking():
 def test_bias_relu_bwd_db_matches_reference():
     import tf_yarn_amd.ops._C as C
     torch.manual_seed(10)
-    for dtype in (torch.float32, torch.bfloat16):
-        y = torch.relu(torch.randn(1024, 516)).cuda().to(dtype)
-        dy = torch.randn(1024, 516).cuda().to(dtype)
+    # cols=516 exercises the quad kernel, 512/1024 the bf16 oct kernel
+    for dtype, cols in [(torch.float32, 516), (torch.bfloat16, 516),
+                        (torch.bfloat16, 512), (torch.bfloat16, 1024)]:
+        y = torch.relu(torch.randn(1024, cols)).cuda().to(dtype)
+        dy = torch.randn(1024, cols).cuda().to(dtype)
         dx, dbias = C.bias_relu_bwd_db(dy.contiguous(), y.contiguous())
         ref_dx = dy.float() * (y.float() > 0)
         assert torch.allclose(dx.float(), ref_dx.to(dtype).float(),
