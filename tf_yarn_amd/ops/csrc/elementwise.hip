@@ -130,61 +130,58 @@ __global__ void bias_relu_bwd_dbpart_kernel(
 // bf16 oct variant of the dbpart kernel: b128 loads/stores (16 B/lane,
 // 8 bf16) instead of b64 quads — the quad version streams at ~4 TB/s,
 // leaving HBM bandwidth on the table for these 2-read/1-write layers.
+// 2D thread map: each thread owns exactly ONE column-oct and the block
+// covers blockDim/octs row slots concurrently, so narrow layers (the
+// 256-wide MLP tail ran at 1.9 TB/s with a 1D map) still fill the chip.
+// Requires octs to divide blockDim; one partial row per (block, slot).
 // dx keeps the original dy bits where the mask passes (no round trip).
-#define DB8_MAX_K 4  // max column-octs per thread (cols <= 8*block*K)
-
 __global__ void bias_relu_bwd_dbpart8_kernel(
     const unsigned short* __restrict__ dy,
     const unsigned short* __restrict__ y,
     unsigned short* __restrict__ dx,
-    float* __restrict__ dbias_part,
+    float* __restrict__ dbias_part,  // [gridDim.x * rpb][cols]
     int64_t rows, int64_t cols) {
-  const int64_t octs = cols >> 3;
-  float acc[DB8_MAX_K][8];
+  const int octs = (int)(cols >> 3);
+  const int rpb = blockDim.x / octs;       // row slots per block
+  const int rid = threadIdx.x / octs;      // my slot
+  const int64_t q = threadIdx.x - rid * octs;  // my oct (fixed)
+  float acc[8];
 #pragma unroll
-  for (int k = 0; k < DB8_MAX_K; ++k)
-#pragma unroll
-    for (int j = 0; j < 8; ++j) acc[k][j] = 0.f;
-  const int64_t row_stride = (int64_t)gridDim.x * DB_ROWS;
-  for (int64_t r0 = (int64_t)blockIdx.x * DB_ROWS; r0 < rows;
-       r0 += row_stride) {
+  for (int j = 0; j < 8; ++j) acc[j] = 0.f;
+  const int64_t slot = (int64_t)blockIdx.x * rpb + rid;
+  const int64_t row_stride = (int64_t)gridDim.x * rpb * DB_ROWS;
+  for (int64_t r0 = slot * DB_ROWS; r0 < rows; r0 += row_stride) {
     const int nr = min((int64_t)DB_ROWS, rows - r0);
-    int k = 0;
-    for (int64_t q = threadIdx.x; q < octs; q += blockDim.x, ++k) {
-      bf16x8 gv[DB_ROWS], yv[DB_ROWS];
+    bf16x8 gv[DB_ROWS], yv[DB_ROWS];
 #pragma unroll
-      for (int rr = 0; rr < DB_ROWS; ++rr)
-        if (rr < nr)
-          gv[rr] = reinterpret_cast<const bf16x8*>(
-              dy)[(r0 + rr) * octs + q];
+    for (int rr = 0; rr < DB_ROWS; ++rr)
+      if (rr < nr)
+        gv[rr] = reinterpret_cast<const bf16x8*>(
+            dy)[(r0 + rr) * octs + q];
 #pragma unroll
-      for (int rr = 0; rr < DB_ROWS; ++rr)
-        if (rr < nr)
-          yv[rr] = reinterpret_cast<const bf16x8*>(
-              y)[(r0 + rr) * octs + q];
+    for (int rr = 0; rr < DB_ROWS; ++rr)
+      if (rr < nr)
+        yv[rr] = reinterpret_cast<const bf16x8*>(
+            y)[(r0 + rr) * octs + q];
 #pragma unroll
-      for (int rr = 0; rr < DB_ROWS; ++rr)
-        if (rr < nr) {
-          bf16x8 out;
+    for (int rr = 0; rr < DB_ROWS; ++rr)
+      if (rr < nr) {
+        bf16x8 out;
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const bool keep = bf16_to_f32(yv[rr][j]) > 0.f;
-            out[j] = keep ? gv[rr][j] : (unsigned short)0;
-            if (keep) acc[k][j] += bf16_to_f32(gv[rr][j]);
-          }
-          reinterpret_cast<bf16x8*>(dx)[(r0 + rr) * octs + q] = out;
+        for (int j = 0; j < 8; ++j) {
+          const bool keep = bf16_to_f32(yv[rr][j]) > 0.f;
+          out[j] = keep ? gv[rr][j] : (unsigned short)0;
+          if (keep) acc[j] += bf16_to_f32(gv[rr][j]);
         }
-    }
+        reinterpret_cast<bf16x8*>(dx)[(r0 + rr) * octs + q] = out;
+      }
   }
-  float* part = dbias_part + (int64_t)blockIdx.x * cols;
-  int k = 0;
-  for (int64_t q = threadIdx.x; q < octs; q += blockDim.x, ++k) {
-    f32x4 lo, hi;
+  float* part = dbias_part + slot * cols;
+  f32x4 lo, hi;
 #pragma unroll
-    for (int j = 0; j < 4; ++j) { lo[j] = acc[k][j]; hi[j] = acc[k][4 + j]; }
-    reinterpret_cast<f32x4*>(part)[q * 2] = lo;
-    reinterpret_cast<f32x4*>(part)[q * 2 + 1] = hi;
-  }
+  for (int j = 0; j < 4; ++j) { lo[j] = acc[j]; hi[j] = acc[4 + j]; }
+  reinterpret_cast<f32x4*>(part)[q * 2] = lo;
+  reinterpret_cast<f32x4*>(part)[q * 2 + 1] = hi;
 }
 
 // dx = dy * (y > 0)
@@ -349,26 +346,31 @@ std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
       std::min<int64_t>(MIYARN_BLOCK, ((quads + 63) / 64) * 64));
   int grid = static_cast<int>(std::min<int64_t>(
       (rows + DB_ROWS - 1) / DB_ROWS, MIYARN_MAX_BLOCKS));
-  // Atomic-free column partials: one row per block, reduced below.
-  auto part = torch::empty({grid, cols},
+  const int64_t octs = cols / 8;
+  const bool oct_ok = dy.scalar_type() == torch::kBFloat16 &&
+                      cols % 8 == 0 && octs <= MIYARN_BLOCK &&
+                      MIYARN_BLOCK % octs == 0;
+  const int64_t rpb = oct_ok ? MIYARN_BLOCK / octs : 1;
+  if (oct_ok)
+    grid = static_cast<int>(std::min<int64_t>(
+        (rows + rpb * DB_ROWS - 1) / (rpb * DB_ROWS), MIYARN_MAX_BLOCKS));
+  // Atomic-free column partials: one row per block (per slot for the
+  // oct kernel), reduced below.
+  auto part = torch::empty({grid * rpb, cols},
                            dy.options().dtype(torch::kFloat32));
-  if (dy.scalar_type() == torch::kFloat32) {
+  if (oct_ok) {
+    hipLaunchKernelGGL(bias_relu_bwd_dbpart8_kernel, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream,
+                       reinterpret_cast<unsigned short*>(dy.data_ptr()),
+                       reinterpret_cast<unsigned short*>(y.data_ptr()),
+                       reinterpret_cast<unsigned short*>(dx.data_ptr()),
+                       part.data_ptr<float>(), rows, cols);
+  } else if (dy.scalar_type() == torch::kFloat32) {
     hipLaunchKernelGGL(bias_relu_bwd_dbpart_kernel<F32Io>, dim3(grid),
                        dim3(block), 0, stream,
                        dy.data_ptr<float>(), y.data_ptr<float>(),
                        dx.data_ptr<float>(), part.data_ptr<float>(),
                        rows, cols);
-  } else if (cols % 8 == 0 && cols <= 8 * MIYARN_BLOCK * DB8_MAX_K) {
-    TORCH_CHECK(dy.scalar_type() == torch::kBFloat16, "fp32/bf16 only");
-    const int64_t octs = cols / 8;
-    int block8 = static_cast<int>(
-        std::min<int64_t>(MIYARN_BLOCK, ((octs + 63) / 64) * 64));
-    hipLaunchKernelGGL(bias_relu_bwd_dbpart8_kernel, dim3(grid),
-                       dim3(block8), 0, stream,
-                       reinterpret_cast<unsigned short*>(dy.data_ptr()),
-                       reinterpret_cast<unsigned short*>(y.data_ptr()),
-                       reinterpret_cast<unsigned short*>(dx.data_ptr()),
-                       part.data_ptr<float>(), rows, cols);
   } else {
     TORCH_CHECK(dy.scalar_type() == torch::kBFloat16, "fp32/bf16 only");
     hipLaunchKernelGGL(bias_relu_bwd_dbpart_kernel<Bf16Io>, dim3(grid),
