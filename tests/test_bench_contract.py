@@ -72,3 +72,18 @@ def test_bench_under_torchrun_4_ranks_uneven_shards():
         cwd=REPO, capture_output=True, text=True, timeout=280)
     assert out.returncode == 0, out.stderr[-2000:]
     _check_json_line(out.stdout, 4)
+
+
+@pytest.mark.timeout(300)
+def test_bench_under_torchrun_8_ranks():
+    """dp8: the exact rank count of the driver's round-end scale run
+    (26 features shard 4,4,3,3,3,3,3,3 over 8 ranks)."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", "29587", "bench.py", "--gpus", "8",
+         "--steps", "2", "--warmup", "1", "--batch", "16",
+         "--table-rows", "200", "--dtype", "fp32"],
+        cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert out.returncode == 0, out.stderr[-2000:]
+    _check_json_line(out.stdout, 8)
