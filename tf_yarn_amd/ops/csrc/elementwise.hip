@@ -350,7 +350,11 @@ std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
   const bool oct_ok = dy.scalar_type() == torch::kBFloat16 &&
                       cols % 8 == 0 && octs <= MIYARN_BLOCK &&
                       MIYARN_BLOCK % octs == 0;
-  const int64_t rpb = oct_ok ? MIYARN_BLOCK / octs : 1;
+  // Wide layers measured best with one row slot per block (block = octs);
+  // narrow ones need multiple slots to fill the chip (65536x256 ran at
+  // 1.9 TB/s single-slot vs 2.2 multi-slot).
+  const int block8 = octs >= 128 ? static_cast<int>(octs) : MIYARN_BLOCK;
+  const int64_t rpb = oct_ok ? block8 / octs : 1;
   if (oct_ok)
     grid = static_cast<int>(std::min<int64_t>(
         (rows + rpb * DB_ROWS - 1) / (rpb * DB_ROWS), MIYARN_MAX_BLOCKS));
@@ -360,7 +364,7 @@ std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
                            dy.options().dtype(torch::kFloat32));
   if (oct_ok) {
     hipLaunchKernelGGL(bias_relu_bwd_dbpart8_kernel, dim3(grid),
-                       dim3(MIYARN_BLOCK), 0, stream,
+                       dim3(block8), 0, stream,
                        reinterpret_cast<unsigned short*>(dy.data_ptr()),
                        reinterpret_cast<unsigned short*>(y.data_ptr()),
                        reinterpret_cast<unsigned short*>(dx.data_ptr()),
