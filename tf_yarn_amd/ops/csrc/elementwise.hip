@@ -356,8 +356,11 @@ std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
   const int block8 = octs >= 128 ? static_cast<int>(octs) : MIYARN_BLOCK;
   const int64_t rpb = oct_ok ? block8 / octs : 1;
   if (oct_ok)
+    // cap grid*rpb (= partial rows) at MAX_BLOCKS so the host-side
+    // partial reduction stays small; slots grid-stride over rows
     grid = static_cast<int>(std::min<int64_t>(
-        (rows + rpb * DB_ROWS - 1) / (rpb * DB_ROWS), MIYARN_MAX_BLOCKS));
+        (rows + rpb * DB_ROWS - 1) / (rpb * DB_ROWS),
+        MIYARN_MAX_BLOCKS / rpb));
   // Atomic-free column partials: one row per block (per slot for the
   // oct kernel), reduced below.
   auto part = torch::empty({grid * rpb, cols},
@@ -422,24 +425,33 @@ template <typename Io>
 __global__ void col_reduce_dot_kernel(
     const typename Io::scalar_t* __restrict__ x,
     const typename Io::scalar_t* __restrict__ dy,
-    float* __restrict__ part,  // [gridDim.x][cols]
+    float* __restrict__ part,  // [gridDim.x * slots][cols]
     int64_t rows, int64_t cols) {
   const int64_t quads = cols >> 2;
+  // 2D map for narrow inputs: when the quad count divides the block,
+  // the block covers blockDim/quads row slots concurrently (a 64-thread
+  // 1D map left the chip underfilled on the 256-wide head).
+  const int bd = blockDim.x;
+  const bool multi = quads <= bd && bd % (int)quads == 0;
+  const int tpq = multi ? (int)quads : bd;
+  const int slots = multi ? bd / (int)quads : 1;
+  const int rid = threadIdx.x / tpq;
+  const int ltid = threadIdx.x - rid * tpq;
   float acc[DB_MAX_K][4];
 #pragma unroll
   for (int k = 0; k < DB_MAX_K; ++k)
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[k][j] = 0.f;
-  const int64_t row_stride = (int64_t)gridDim.x * DB_ROWS;
-  for (int64_t r0 = (int64_t)blockIdx.x * DB_ROWS; r0 < rows;
-       r0 += row_stride) {
+  const int64_t slot = (int64_t)blockIdx.x * slots + rid;
+  const int64_t row_stride = (int64_t)gridDim.x * slots * DB_ROWS;
+  for (int64_t r0 = slot * DB_ROWS; r0 < rows; r0 += row_stride) {
     const int nr = min((int64_t)DB_ROWS, rows - r0);
     float w[DB_ROWS];
 #pragma unroll
     for (int rr = 0; rr < DB_ROWS; ++rr)
       w[rr] = rr < nr ? Io::load(dy, r0 + rr) : 0.f;
     int k = 0;
-    for (int64_t q = threadIdx.x; q < quads; q += blockDim.x, ++k) {
+    for (int64_t q = ltid; q < quads; q += tpq, ++k) {
       float v[DB_ROWS][4];
 #pragma unroll
       for (int rr = 0; rr < DB_ROWS; ++rr)
@@ -451,9 +463,9 @@ __global__ void col_reduce_dot_kernel(
           for (int j = 0; j < 4; ++j) acc[k][j] += w[rr] * v[rr][j];
     }
   }
-  float* prow = part + (int64_t)blockIdx.x * cols;
+  float* prow = part + slot * cols;
   int k = 0;
-  for (int64_t q = threadIdx.x; q < quads; q += blockDim.x, ++k) {
+  for (int64_t q = ltid; q < quads; q += tpq, ++k) {
     f32x4 o;
 #pragma unroll
     for (int j = 0; j < 4; ++j) o[j] = acc[k][j];
@@ -475,11 +487,15 @@ torch::Tensor col_reduce_dot(torch::Tensor x, torch::Tensor dy) {
               "col_reduce_dot needs cols % 4 == 0 and small cols");
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int64_t quads = cols / 4;
-  int block = static_cast<int>(
-      std::min<int64_t>(MIYARN_BLOCK, ((quads + 63) / 64) * 64));
-  int grid = static_cast<int>(std::min<int64_t>(
-      (rows + DB_ROWS - 1) / DB_ROWS, MIYARN_MAX_BLOCKS));
-  auto part = torch::empty({grid, cols},
+  const bool multi = quads <= MIYARN_BLOCK && MIYARN_BLOCK % quads == 0;
+  int block = multi ? MIYARN_BLOCK
+                    : static_cast<int>(std::min<int64_t>(
+                          MIYARN_BLOCK, ((quads + 63) / 64) * 64));
+  const int64_t slots = multi ? MIYARN_BLOCK / quads : 1;
+  int grid = static_cast<int>(std::max<int64_t>(
+      1, std::min<int64_t>((rows + slots * DB_ROWS - 1) / (slots * DB_ROWS),
+                           MIYARN_MAX_BLOCKS / slots)));
+  auto part = torch::empty({grid * slots, cols},
                            x.options().dtype(torch::kFloat32));
   if (x.scalar_type() == torch::kFloat32) {
     hipLaunchKernelGGL(col_reduce_dot_kernel<F32Io>, dim3(grid),
