@@ -309,9 +309,29 @@ class ShardedCriteoEmbeddings(nn.Module):
         return out_buf, wide
 
     @torch.no_grad()
-    def apply_sparse_updates(self, lr: float) -> None:
+    def apply_sparse_updates(self, lr: float, stream=None) -> None:
         """Owner-local fused scatter+SGD; scale 1/world (grads come from a
-        global-mean loss split across ranks)."""
+        global-mean loss split across ranks).
+
+        ``stream``: optional side HIP stream.  The scatters are atomic-
+        bound (~0.8 TB/s, CUs mostly idle), so running them concurrently
+        with the dense optimizer step hides most of their latency; the
+        CALLER must make the compute stream wait on ``stream`` before the
+        next forward's gather reads the tables."""
+        if stream is not None:
+            stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(stream):
+                for flat_ids, grad in self._deep_sink:
+                    flat_ids.record_stream(stream)
+                    grad.record_stream(stream)
+                for flat_ids, gw in self._wide_sink:
+                    flat_ids.record_stream(stream)
+                    gw.record_stream(stream)
+                self._apply_pending(lr)
+            return
+        self._apply_pending(lr)
+
+    def _apply_pending(self, lr: float) -> None:
         scale = 1.0 / self.world
         for flat_ids, grad in self._deep_sink:
             ops.emb_bwd_sgd(self.weight.data, flat_ids,

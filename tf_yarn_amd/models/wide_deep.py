@@ -403,9 +403,9 @@ class WideAndDeep(nn.Module):
         self.deep_embedding.start_sparse_sync(process_group)
         self.wide_embedding.start_sparse_sync(process_group)
 
-    def finish_sparse_sync(self, lr: float) -> None:
+    def finish_sparse_sync(self, lr: float, stream=None) -> None:
         if self.sharded:
-            self.embeddings.apply_sparse_updates(lr)
+            self.embeddings.apply_sparse_updates(lr, stream=stream)
             return
         self.deep_embedding.finish_sparse_sync(lr)
         self.wide_embedding.finish_sparse_sync(lr)
