@@ -112,7 +112,6 @@ def main() -> None:
     lr = 0.02
     opt = FusedSGD([p for p in model.parameters()
                     if not getattr(p, "_miyarn_sparse", False)], lr=lr)
-    loss_fn = torch.nn.BCEWithLogitsLoss()
 
     # Pre-generate distinct synthetic batches on-device (data=synthetic;
     # varying ids each step so the gather/scatter path is exercised).
@@ -134,8 +133,9 @@ def main() -> None:
     def eager_step(i: int) -> torch.Tensor:
         dense, ids, labels = batches[i % N_DATA_BATCHES]
         opt.zero_grad(set_to_none=set_to_none)
-        logits = ddp(dense, ids)
-        loss = loss_fn(logits.float(), labels)
+        # labels passed through: the 3-part logit sum is fused into the
+        # BCE loss kernel (ops.bce_head_loss)
+        loss = ddp(dense, ids, labels=labels)
         loss.backward()
         # sparse exchange overlaps with the dense optimizer step
         module.start_sparse_sync()

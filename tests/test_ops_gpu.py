@@ -592,3 +592,23 @@ def test_scalar_head_ragged_width_padded_path():
     (x @ head.weight + head.bias).float().sum().backward()
     assert torch.allclose(gw.float(), head.weight.grad.float(),
                           atol=0.05, rtol=0.05)
+
+
+@requires_gpu
+def test_bce_head_fused_gpu():
+    """Fused 3-part BCE head vs torch reference, forward and grads."""
+    torch.manual_seed(26)
+    n = 4096
+    parts = [(torch.randn(n, device="cuda") / 4).to(torch.bfloat16)
+             .requires_grad_(True) for _ in range(3)]
+    labels = (torch.rand(n, device="cuda") > 0.5).float()
+    loss = ops.bce_head_loss(*parts, labels)
+    loss.backward()
+    parts2 = [p.detach().clone().requires_grad_(True) for p in parts]
+    z = (parts2[0].float() + parts2[1].float() + parts2[2].float())
+    ref = torch.nn.functional.binary_cross_entropy_with_logits(z, labels)
+    ref.backward()
+    assert torch.allclose(loss, ref, atol=2e-3), (loss, ref)
+    for p, p2 in zip(parts, parts2):
+        assert torch.allclose(p.grad.float(), p2.grad.float(),
+                              atol=1e-4), "grad mismatch"

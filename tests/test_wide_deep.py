@@ -91,3 +91,27 @@ def test_resnet_forward_backward_cpu():
     # resnet50 has the canonical parameter count (~25.6M)
     n_params = sum(p.numel() for p in resnet50().parameters())
     assert abs(n_params - 25_557_032) < 1000, n_params
+
+
+def test_forward_with_labels_matches_separate_loss():
+    """model(dense, ids, labels=...) (fused BCE head) must equal
+    BCEWithLogits over the plain logits path."""
+    from tf_yarn_amd.models.synthetic import synthetic_criteo_batch
+    from tf_yarn_amd.models.wide_deep import WideAndDeep
+
+    torch.manual_seed(11)
+    tables = [50] * 26
+    for sharded in (False, True):
+        model = WideAndDeep(table_sizes=tables, embedding_dim=4,
+                            hidden=(16, 8), sharded=sharded)
+        dense, ids, labels = synthetic_criteo_batch(32, tables, seed=3)
+        logits = model(dense, ids)
+        ref = torch.nn.functional.binary_cross_entropy_with_logits(
+            logits.float(), labels)
+        model.clear_pending()
+        fused = model(dense, ids, labels=labels)
+        model.clear_pending()
+        assert torch.allclose(fused, ref, atol=1e-5), (sharded, fused, ref)
+        fused.backward()  # grads flow through all three parts
+        assert model.wide_dense.weight.grad is not None
+        assert model.head.weight.grad is not None

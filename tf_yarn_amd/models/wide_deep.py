@@ -368,8 +368,11 @@ class WideAndDeep(nn.Module):
             # dense compute in bf16; masters are managed by the optimizer
             self.mlp = self.mlp.to(torch.bfloat16)
 
-    def forward(self, dense: torch.Tensor,
-                sparse_ids: torch.Tensor) -> torch.Tensor:
+    def forward(self, dense: torch.Tensor, sparse_ids: torch.Tensor,
+                labels: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Returns logits [B], or — when ``labels`` is given — the mean
+        BCEWithLogits loss with the 3-part logit sum fused into the loss
+        kernel (ops.bce_head_loss)."""
         dense = dense.to(self.compute_dtype)
         if self.sharded:
             b = dense.shape[0]
@@ -382,16 +385,22 @@ class WideAndDeep(nn.Module):
             deep_in, wide_sum = emb(sparse_ids, deep_in,
                                     _DeepInput.DENSE_PAD)
             deep_out = self.head(self.mlp(deep_in))
-            return deep_out + wide_sum.to(deep_out.dtype) \
-                + self.wide_dense(dense)
+            wide_sum = wide_sum.to(deep_out.dtype)
+            wd = self.wide_dense(dense)
+            if labels is not None:
+                return ops.bce_head_loss(deep_out, wide_sum, wd, labels)
+            return deep_out + wide_sum + wd
         emb = self.deep_embedding
         flat = (sparse_ids + emb.offsets.unsqueeze(0)).reshape(-1)
         deep_in = _DeepInput.apply(
             dense, emb.weight, flat, emb.dim,
             self.compute_dtype == torch.bfloat16, emb._sink)
         deep_out = self.head(self.mlp(deep_in))
-        wide_out = self.wide_embedding(sparse_ids) + self.wide_dense(dense)
-        return deep_out + wide_out
+        wide_emb = self.wide_embedding(sparse_ids).to(deep_out.dtype)
+        wd = self.wide_dense(dense)
+        if labels is not None:
+            return ops.bce_head_loss(deep_out, wide_emb, wd, labels)
+        return deep_out + wide_emb + wd
 
     def start_sparse_sync(self, process_group=None) -> None:
         """Kick off the replicated-mode allgathers (called right after

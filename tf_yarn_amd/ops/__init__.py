@@ -323,6 +323,46 @@ class ScalarHeadFn(torch.autograd.Function):
         return dx, dw, db
 
 
+class BCEHeadFn(torch.autograd.Function):
+    """Per-example BCEWithLogits over logits = deep + wide + dhead,
+    fused into one kernel pair (saves the two adds, the fp32 cast, and
+    torch's separate loss fwd/bwd elementwise kernels)."""
+
+    @staticmethod
+    def forward(ctx, deep, wide, dhead, labels):
+        if deep.is_cuda and HAVE_EXT and deep.dtype == torch.bfloat16 \
+                and wide.dtype == torch.bfloat16 \
+                and dhead.dtype == torch.bfloat16:
+            loss, sig = _C.bce_head_fwd(
+                deep.contiguous(), wide.contiguous(), dhead.contiguous(),
+                labels.contiguous().float())
+        else:
+            z = (deep.float() + wide.float() + dhead.float())
+            labels = labels.float()
+            loss = torch.nn.functional.binary_cross_entropy_with_logits(
+                z, labels, reduction="none")
+            sig = torch.sigmoid(z).to(deep.dtype)
+        ctx.save_for_backward(sig, labels)
+        ctx.dtype = deep.dtype
+        return loss
+
+    @staticmethod
+    def backward(ctx, g):
+        sig, labels = ctx.saved_tensors
+        if sig.is_cuda and HAVE_EXT and sig.dtype == torch.bfloat16:
+            dl = _C.bce_head_bwd(sig, labels.contiguous().float(),
+                                 g.contiguous().float())
+        else:
+            dl = ((sig.float() - labels.float()) * g.float()).to(ctx.dtype)
+        return dl, dl, dl, None
+
+
+def bce_head_loss(deep: torch.Tensor, wide: torch.Tensor,
+                  dhead: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
+    """Mean BCEWithLogits over the fused 3-part head."""
+    return BCEHeadFn.apply(deep, wide, dhead, labels).mean()
+
+
 class BiasReLU(torch.autograd.Function):
     """Autograd wrapper for the fused bias+ReLU epilogue (backward fuses
     the dbias reduction into the dx kernel on GPU)."""

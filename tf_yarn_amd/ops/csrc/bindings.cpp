@@ -54,6 +54,12 @@ std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
 torch::Tensor col_reduce_dot(torch::Tensor x, torch::Tensor dy);
 torch::Tensor row_dot(torch::Tensor x, torch::Tensor w,
                       c10::optional<torch::Tensor> bias);
+std::vector<torch::Tensor> bce_head_fwd(torch::Tensor deep,
+                                        torch::Tensor wide,
+                                        torch::Tensor dhead,
+                                        torch::Tensor labels);
+torch::Tensor bce_head_bwd(torch::Tensor sig, torch::Tensor labels,
+                           torch::Tensor g);
 
 // wgrad.hip
 torch::Tensor wgrad_nt(torch::Tensor dy, torch::Tensor x, int64_t splitk);
@@ -92,6 +98,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "dw[m] = sum_b dy[b] * x[b,m] (single-logit head wgrad)");
   m.def("row_dot", &row_dot,
         "y[b] = x[b].w + bias (single-logit head forward GEMV)");
+  m.def("bce_head_fwd", &bce_head_fwd,
+        "Fused 3-part logit sum + stable BCEWithLogits (loss, sigmoid)");
+  m.def("bce_head_bwd", &bce_head_bwd,
+        "dlogit = (sigmoid - label) * upstream grad");
   m.def("bias_relu_bwd_db", &bias_relu_bwd_db,
         "Fused ReLU backward + dbias reduction (returns [dx, dbias_fp32])");
   m.def("emb_fwd_into", &emb_fwd_into,
