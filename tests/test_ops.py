@@ -154,3 +154,62 @@ def test_gemm_bt_cpu_fallback():
     assert torch.allclose(out, ref, atol=1e-5)
     out2 = ops.gemm_bt(a, b, None, relu=False)
     assert torch.allclose(out2, a @ b.t(), atol=1e-5)
+
+
+def test_row_dot_cpu_fallback():
+    torch.manual_seed(30)
+    x = torch.randn(64, 12)
+    w = torch.randn(12)
+    b = torch.randn(1)
+    assert torch.allclose(ops.row_dot(x, w, b), x @ w + b, atol=1e-5)
+    assert torch.allclose(ops.row_dot(x, w, None), x @ w, atol=1e-5)
+
+
+def test_bce_head_loss_cpu_matches_torch():
+    torch.manual_seed(31)
+    parts = [torch.randn(40, requires_grad=True) for _ in range(3)]
+    labels = (torch.rand(40) > 0.5).float()
+    loss = ops.bce_head_loss(*parts, labels)
+    ref_parts = [p.detach().clone().requires_grad_(True) for p in parts]
+    ref = torch.nn.functional.binary_cross_entropy_with_logits(
+        ref_parts[0] + ref_parts[1] + ref_parts[2], labels)
+    assert torch.allclose(loss, ref, atol=1e-6)
+    loss.backward()
+    ref.backward()
+    for p, rp in zip(parts, ref_parts):
+        assert torch.allclose(p.grad, rp.grad, atol=1e-6)
+
+
+def test_emb_fwd_into_and_gather_scatter_cpu():
+    torch.manual_seed(32)
+    table = torch.randn(20, 4)
+    ids = torch.randint(0, 20, (6, 3))
+    out = torch.zeros(6, 2 + 12)
+    ops.emb_fwd_into(table, ids.reshape(-1), out, col_offset=2)
+    ref = table.index_select(0, ids.reshape(-1)).reshape(6, 12)
+    assert torch.allclose(out[:, 2:], ref)
+    # scalar wide table
+    wide = torch.randn(20, 1)
+    s = ops.emb_gather_sum(wide, ids)
+    ref_s = wide.reshape(-1).index_select(0, ids.reshape(-1)) \
+        .reshape(6, 3).sum(dim=1)
+    assert torch.allclose(s, ref_s, atol=1e-6)
+    wide2 = wide.clone()
+    g = torch.randn(6)
+    ops.emb_scatter_sum(wide2, ids, g, alpha=-0.1)
+    ref_w = wide.reshape(-1).clone()
+    for b in range(6):
+        for f in range(3):
+            ref_w[ids[b, f]] += -0.1 * g[b]
+    assert torch.allclose(wide2.reshape(-1), ref_w, atol=1e-5)
+
+
+def test_emb_bwd_dense_cpu():
+    torch.manual_seed(33)
+    gt = torch.zeros(10, 4)
+    ids = torch.tensor([1, 1, 3])
+    grad = torch.randn(3, 4)
+    ops.emb_bwd_dense(gt, ids, grad, scale=0.5)
+    ref = torch.zeros(10, 4)
+    ref.index_add_(0, ids, grad * 0.5)
+    assert torch.allclose(gt, ref, atol=1e-6)
