@@ -87,3 +87,23 @@ def test_bench_under_torchrun_8_ranks():
         cwd=REPO, capture_output=True, text=True, timeout=280)
     assert out.returncode == 0, out.stderr[-2000:]
     _check_json_line(out.stdout, 8)
+
+
+@pytest.mark.timeout(300)
+def test_bench_ps_contract(tmp_path):
+    """scripts/bench_ps.py (BASELINE config 2): full chief+1ps+2workers
+    topology end-to-end, one JSON line with the shared field set."""
+    out = subprocess.run(
+        [sys.executable, "scripts/bench_ps.py", "--steps", "40",
+         "--batch", "256"],
+        cwd=REPO, capture_output=True, text=True, timeout=280,
+        env=dict(os.environ, MODEL_DIR=str(tmp_path / "ps_model")))
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines()
+             if l.startswith("{")]
+    assert len(lines) == 1, out.stdout[-1000:]
+    result = json.loads(lines[0])
+    for field in REQUIRED_FIELDS:
+        assert field in result, f"missing field {field}"
+    assert result["value"] > 0
+    assert result["config"]["parallelism"] == "ps-async"
