@@ -158,3 +158,27 @@ def test_step_per_second_hook_counts(tmp_path):
     hook = StepPerSecondHook(every_n_steps=2)
     est.train(_input_fn, max_steps=6, hooks=[hook])
     assert est.global_step == 6
+
+
+def test_linear_classifier_train_eval(tmp_path):
+    """Premade LinearClassifier (reference linear_classifier_example.py
+    family): learns a separable problem, writes model.ckpt-N."""
+    import torch
+
+    from tf_yarn_amd.estimator import LinearClassifier, RunConfig
+
+    est = LinearClassifier(
+        n_features=6, model_dir=str(tmp_path / "m"),
+        config=RunConfig(save_checkpoints_steps=20))
+
+    def input_fn():
+        torch.manual_seed(4)
+        w = torch.randn(6)
+        for _ in range(40):
+            x = torch.randn(64, 6)
+            yield x, ((x @ w) > 0).long()
+
+    est.train(input_fn, max_steps=80)
+    result = est.evaluate(input_fn, steps=10)
+    assert result["accuracy"] > 0.9, result
+    assert est.latest_checkpoint().endswith("model.ckpt-80")

@@ -1,6 +1,7 @@
 from tf_yarn_amd.estimator.client import run_on_yarn
 from tf_yarn_amd.estimator.estimator import (DNNClassifier, Estimator,
-                                             EvalSpec, RunConfig, TrainSpec,
+                                             EvalSpec, LinearClassifier,
+                                             RunConfig, TrainSpec,
                                              train_and_evaluate)
 from tf_yarn_amd.estimator.experiment import Experiment
 from tf_yarn_amd.estimator.keras import (KerasModel, ModelCheckpoint,
@@ -8,6 +9,7 @@ from tf_yarn_amd.estimator.keras import (KerasModel, ModelCheckpoint,
 from tf_yarn_amd.estimator.keras_experiment import KerasExperiment
 
 __all__ = ["run_on_yarn", "Experiment", "KerasExperiment",
-           "Estimator", "DNNClassifier", "TrainSpec", "EvalSpec",
+           "Estimator", "DNNClassifier", "LinearClassifier",
+           "TrainSpec", "EvalSpec",
            "RunConfig", "train_and_evaluate", "KerasModel",
            "ModelCheckpoint", "load_model"]

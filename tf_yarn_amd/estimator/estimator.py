@@ -347,6 +347,35 @@ def train_and_evaluate(estimator: Estimator, train_spec: TrainSpec,
                            name=eval_spec.name)
 
 
+class LinearClassifier(Estimator):
+    """Premade linear classifier (the reference's
+    ``examples/linear_classifier_example.py`` uses
+    ``tf.estimator.LinearClassifier`` over winequality)."""
+
+    def __init__(self, n_features: int, n_classes: int = 2,
+                 optimizer_fn: Optional[Callable] = None,
+                 model_dir: Optional[str] = None,
+                 config: Optional[RunConfig] = None,
+                 device: Optional[str] = None):
+        def module_fn() -> nn.Module:
+            return nn.Linear(n_features, n_classes)
+
+        def loss_fn(outputs, labels):
+            return nn.functional.cross_entropy(outputs, labels.long())
+
+        def metrics_fn(outputs, labels):
+            acc = (outputs.argmax(dim=1) == labels.long()).float().mean()
+            return {"accuracy": float(acc)}
+
+        if optimizer_fn is None:
+            def optimizer_fn(params):
+                return torch.optim.SGD(params, lr=0.1)
+
+        super().__init__(module_fn, optimizer_fn, loss_fn,
+                         model_dir=model_dir, config=config,
+                         metrics_fn=metrics_fn, device=device)
+
+
 class DNNClassifier(Estimator):
     """Premade DNN classifier (the reference's Estimator examples use
     ``tf.estimator.DNNClassifier`` over winequality/Criteo tabular data)."""
