@@ -13,11 +13,14 @@ Submodules (``client``, ``experiment``, ``keras_experiment``, ``metrics``,
 
 import sys
 
+from tf_yarn_amd import (Metrics, NodeLabel, RunFailed, TaskSpec,
+                         get_safe_experiment_fn, ps_strategy_topology,
+                         single_server_topology)
 from tf_yarn_amd.estimator import (DNNClassifier, Estimator, EvalSpec,
                                    Experiment, KerasExperiment, KerasModel,
-                                   ModelCheckpoint, RunConfig, TrainSpec,
-                                   load_model, run_on_yarn,
-                                   train_and_evaluate)
+                                   LinearClassifier, ModelCheckpoint,
+                                   RunConfig, TrainSpec, load_model,
+                                   run_on_yarn, train_and_evaluate)
 from tf_yarn_amd.estimator import client, cluster, estimator, experiment
 from tf_yarn_amd.estimator import keras_experiment, metrics, tasks
 
@@ -30,6 +33,10 @@ for _name, _mod in [("client", client), ("cluster", cluster),
     sys.modules[f"{__name__}.{_name}"] = _mod
 
 __all__ = ["run_on_yarn", "Experiment", "KerasExperiment", "Estimator",
-           "DNNClassifier", "TrainSpec", "EvalSpec", "RunConfig",
-           "train_and_evaluate", "KerasModel", "ModelCheckpoint",
-           "load_model"]
+           "DNNClassifier", "LinearClassifier", "TrainSpec", "EvalSpec",
+           "RunConfig", "train_and_evaluate", "KerasModel",
+           "ModelCheckpoint", "load_model",
+           # reference tensorflow/__init__.py:1-15 re-exports
+           "RunFailed", "Metrics", "TaskSpec", "NodeLabel",
+           "single_server_topology", "ps_strategy_topology",
+           "get_safe_experiment_fn"]
