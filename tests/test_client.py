@@ -102,7 +102,8 @@ def test_run_on_yarn_retries(nb_retries, nb_failures, expect_raise,
     """Retry semantics (reference tests/test_client.py:165-198)."""
     calls = {"n": 0}
 
-    def fake_setup(task_specs, n_try, custom, hook, base_dir):
+    def fake_setup(task_specs, n_try, custom, hook, base_dir,
+                   extra_env=None):
         return mock.MagicMock()
 
     def fake_execute(cluster, fn, thresholds, n_try):
@@ -142,3 +143,26 @@ def test_tensorflow_alias_package():
     assert run_on_yarn is real_client.run_on_yarn
     from tf_yarn_amd.estimator.experiment import Experiment as E2
     assert Experiment is E2
+
+
+def test_run_on_yarn_env_reaches_tasks(tmp_path):
+    """The env= kwarg (reference client.py:306) must land in every task's
+    environment."""
+    import cloudpickle
+    import sys
+    cloudpickle.register_pickle_by_value(sys.modules[__name__])
+    from tf_yarn_amd import TaskSpec, run_on_yarn
+
+    def experiment_fn():
+        def run():
+            import os
+            assert os.environ["MIYARN_TEST_FLAG"] == "42"
+        return run
+
+    metrics = run_on_yarn(
+        experiment_fn,
+        {"chief": TaskSpec(memory=512, vcores=1)},
+        custom_task_module="tf_yarn_amd.distributed.task",
+        env={"MIYARN_TEST_FLAG": 42},
+        base_dir=str(tmp_path))
+    assert metrics is not None

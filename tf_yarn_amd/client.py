@@ -137,7 +137,9 @@ def _task_env(task: ContainerTask,
               gpus: List[int],
               n_try: int,
               task_specs: TaskSpecs,
-              log_path: str) -> Dict[str, str]:
+              log_path: str,
+              extra_env: Optional[Dict[str, str]] = None
+              ) -> Dict[str, str]:
     """Build the per-container environment contract
     (reference ``client.py:108-133`` + TB_* plumbing ``client.py:213-219``)."""
     child = dict(os.environ)
@@ -170,6 +172,9 @@ def _task_env(task: ContainerTask,
             child["TB_MODEL_DIR"] = spec.tb_model_dir
         if spec.tb_extra_args:
             child["TB_EXTRA_ARGS"] = spec.tb_extra_args
+    # user-supplied env (reference run_on_yarn(env={...}), client.py:306)
+    if extra_env:
+        child.update({k: str(v) for k, v in extra_env.items()})
     return child
 
 
@@ -180,7 +185,9 @@ def _spawn_tasks(tasks: List[ContainerTask],
                  app_dir: str,
                  n_try: int,
                  custom_task_module: Optional[str],
-                 pre_script_hook: Optional[str]) -> List[TaskProcess]:
+                 pre_script_hook: Optional[str],
+                 extra_env: Optional[Dict[str, str]] = None
+                 ) -> List[TaskProcess]:
     gpu_map = _allocate_gpus(tasks, task_specs)
     log_dir = os.path.join(app_dir, "logs")
     os.makedirs(log_dir, exist_ok=True)
@@ -190,7 +197,7 @@ def _spawn_tasks(tasks: List[ContainerTask],
         log_path = os.path.join(log_dir, f"{task.type}_{task.id}.log")
         child_env = _task_env(task, server.address, app_id, app_dir,
                               gpu_map[f"{task.type}:{task.id}"], n_try,
-                              task_specs, log_path)
+                              task_specs, log_path, extra_env)
         cmd = [sys.executable, "-m", module]
         if pre_script_hook:
             cmd = ["bash", "-c",
@@ -224,7 +231,9 @@ def _setup_cluster(task_specs: TaskSpecs,
                    n_try: int,
                    custom_task_module: Optional[str],
                    pre_script_hook: Optional[str],
-                   base_dir: Optional[str] = None) -> LocalCluster:
+                   base_dir: Optional[str] = None,
+                   extra_env: Optional[Dict[str, str]] = None
+                   ) -> LocalCluster:
     """The ``_setup_skein_cluster`` equivalent (reference ``client.py:179``)."""
     app_id = f"miyarn_{uuid.uuid4().hex[:12]}"
     app_dir = os.path.join(base_dir or os.environ.get(
@@ -240,7 +249,8 @@ def _setup_cluster(task_specs: TaskSpecs,
         name="event-listener", daemon=True)
     listener.start()
     processes = _spawn_tasks(tasks, task_specs, server, app_id, app_dir,
-                             n_try, custom_task_module, pre_script_hook)
+                             n_try, custom_task_module, pre_script_hook,
+                             extra_env)
     return LocalCluster(server, client, app_id, app_dir, tasks, processes,
                         listener, events)
 
@@ -400,14 +410,16 @@ def run_on_yarn(experiment_fn,
                 queue: str = "default",
                 ps_strategy: bool = False,
                 base_dir: Optional[str] = None,
+                env: Optional[Dict[str, str]] = None,
                 **_ignored) -> Optional[Metrics]:
     """Launch a distributed experiment on the local MI355X node.
 
     API parity with the reference's ``run_on_yarn`` (``client.py:299-470``):
     same task-spec topology model, cloudpickled ``experiment_fn``, whole-run
     retry loop, event aggregation to :class:`Metrics`, ``RunFailed`` on task
-    failure.  ``name``/``queue`` are accepted for compatibility and unused
-    (there is no YARN queue on one node).
+    failure.  ``env`` adds user variables to every task's environment (reference
+    ``client.py:306``); ``name``/``queue`` are accepted for compatibility
+    and unused (there is no YARN queue on one node).
     """
     if task_specs is None:
         task_specs = {"chief": TaskSpec()}
@@ -425,7 +437,7 @@ def run_on_yarn(experiment_fn,
             with catchtime(f"setting up cluster (try {n_try})"):
                 cluster = _setup_cluster(
                     task_specs, n_try, custom_task_module, pre_script_hook,
-                    base_dir)
+                    base_dir, env)
             try:
                 metrics, container_status = _execute_and_await_termination(
                     cluster, serialized_fn,
