@@ -154,7 +154,7 @@ def test_run_on_yarn_env_reaches_tasks(tmp_path):
     from tf_yarn_amd import TaskSpec, run_on_yarn
 
     def experiment_fn():
-        def run():
+        def run(task_params):
             import os
             assert os.environ["MIYARN_TEST_FLAG"] == "42"
         return run
