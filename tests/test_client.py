@@ -153,11 +153,9 @@ def test_run_on_yarn_env_reaches_tasks(tmp_path):
     cloudpickle.register_pickle_by_value(sys.modules[__name__])
     from tf_yarn_amd import TaskSpec, run_on_yarn
 
-    def experiment_fn():
-        def run(task_params):
-            import os
-            assert os.environ["MIYARN_TEST_FLAG"] == "42"
-        return run
+    def experiment_fn(task_params):
+        import os
+        assert os.environ["MIYARN_TEST_FLAG"] == "42"
 
     metrics = run_on_yarn(
         experiment_fn,
