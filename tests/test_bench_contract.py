@@ -107,3 +107,21 @@ def test_bench_ps_contract(tmp_path):
         assert field in result, f"missing field {field}"
     assert result["value"] > 0
     assert result["config"]["parallelism"] == "ps-async"
+
+
+@pytest.mark.timeout(300)
+def test_bench_resnet_contract():
+    """scripts/bench_resnet.py (BASELINE config 4): same JSON contract."""
+    out = subprocess.run(
+        [sys.executable, "scripts/bench_resnet.py", "--steps", "1",
+         "--warmup", "0", "--batch", "2", "--nchw"],
+        cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines()
+             if l.startswith("{")]
+    assert len(lines) == 1
+    result = json.loads(lines[0])
+    for field in REQUIRED_FIELDS:
+        assert field in result, f"missing field {field}"
+    assert result["config"]["model"] == "resnet50"
+    assert result["value"] > 0
