@@ -67,3 +67,25 @@ def test_ps_topology_constraints():
             "chief": TaskSpec(),
             "ps": TaskSpec(instances=0),
         })
+
+
+def test_memory_accepts_skein_style_strings():
+    """Reference TaskSpec takes Union[int, str] memory ("2 GiB") via
+    skein Resources (topologies.py:64-72)."""
+    assert TaskSpec(memory="2 GiB").memory == 2048
+    assert TaskSpec(memory="512 MiB").memory == 512
+    assert TaskSpec(memory="1 GB").memory == 1024
+    assert TaskSpec(memory="48*1024" if False else 1024).memory == 1024
+    assert TaskSpec(memory=2048).memory == 2048
+    assert TaskSpec(memory="2048").memory == 2048
+
+
+def test_evaluation_only_topology_with_custom_module():
+    """The reference README's evaluation-only flow (README.md:371-380):
+    run_on_yarn never validates custom-module topologies for a chief."""
+    from tf_yarn_amd.topologies import _check_general_topology
+    _check_general_topology({"evaluator": TaskSpec()},
+                            require_chief=False)  # must not raise
+    with pytest.raises(ValueError):
+        _check_general_topology({"evaluator": TaskSpec()},
+                                require_chief=True)

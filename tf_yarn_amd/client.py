@@ -426,7 +426,10 @@ def run_on_yarn(experiment_fn,
     if ps_strategy or "ps" in task_specs:
         _check_ps_topology(task_specs)
     else:
-        _check_general_topology(task_specs)
+        # custom task modules define their own topology needs (e.g. the
+        # reference README's evaluation-only run, README.md:371-380)
+        _check_general_topology(task_specs,
+                                require_chief=custom_task_module is None)
 
     with catchtime("serializing experiment_fn"):
         serialized_fn = cloudpickle.dumps(experiment_fn)

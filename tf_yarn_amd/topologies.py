@@ -55,18 +55,39 @@ class ContainerTask(NamedTuple):
         return ContainerKey(self.type, self.id)
 
 
+def _parse_memory(value) -> int:
+    """Memory in MiB from an int (MiB) or a skein-style string like
+    ``"2 GiB"`` / ``"512 MiB"`` / ``"1 GB"`` (the reference's TaskSpec
+    accepts ``Union[int, str]`` via ``skein.model.Resources``,
+    ``topologies.py:64-72``)."""
+    if isinstance(value, (int, float)):
+        return int(value)
+    text = str(value).strip().lower()
+    units = {"kib": 1 / 1024, "kb": 1 / 1024, "k": 1 / 1024,
+             "mib": 1, "mb": 1, "m": 1,
+             "gib": 1024, "gb": 1024, "g": 1024,
+             "tib": 1024 * 1024, "tb": 1024 * 1024, "t": 1024 * 1024,
+             "b": 1 / (1024 * 1024)}
+    for suffix in sorted(units, key=len, reverse=True):
+        if text.endswith(suffix):
+            number = text[:-len(suffix)].strip()
+            return int(float(number) * units[suffix])
+    return int(float(text))  # bare number string: MiB
+
+
 class TaskSpec:
     """Resources requested for all instances of one task role.
 
     Field names mirror the reference (``topologies.py:54-95``):
-    ``memory`` (MiB), ``vcores``, ``instances``, ``nb_proc_per_worker``,
+    ``memory`` (MiB, or a skein-style string like ``"2 GiB"``),
+    ``vcores``, ``instances``, ``nb_proc_per_worker``,
     ``label``, ``tb_termination_timeout_seconds``, ``tb_model_dir``,
     ``tb_extra_args``.  ``nb_proc_per_worker`` is the number of training
     processes (one per GPU for GPU-labelled tasks) per instance.
     """
 
     def __init__(self,
-                 memory: int = 1024,
+                 memory=1024,
                  vcores: int = 1,
                  instances: int = 1,
                  nb_proc_per_worker: int = 1,
@@ -74,7 +95,7 @@ class TaskSpec:
                  tb_termination_timeout_seconds: int = -1,
                  tb_model_dir: Optional[str] = None,
                  tb_extra_args: Optional[str] = None):
-        self.memory = memory
+        self.memory = _parse_memory(memory)
         self.vcores = vcores
         self.instances = instances
         self.nb_proc_per_worker = nb_proc_per_worker
@@ -93,14 +114,21 @@ class TaskSpec:
 TaskSpecs = Dict[str, TaskSpec]
 
 
-def _check_general_topology(task_specs: TaskSpecs) -> None:
-    """Reference ``topologies.py:97-116``."""
+def _check_general_topology(task_specs: TaskSpecs,
+                            require_chief: bool = True) -> None:
+    """Reference ``topologies.py:97-116``.  ``require_chief=False`` for
+    custom task modules: the reference only validates topologies built by
+    its factory functions, so e.g. the README's evaluation-only flow
+    (``task_specs={"evaluator": ...}, custom_task_module=...``,
+    ``README.md:371-380``) must be accepted."""
     unknown = set(task_specs) - ALL_TASK_TYPES
     if unknown:
         raise ValueError(
             f"unknown task types {sorted(unknown)}; "
             f"supported: {sorted(ALL_TASK_TYPES)}")
-    if "chief" not in task_specs or task_specs["chief"].instances != 1:
+    if require_chief and (
+            "chief" not in task_specs
+            or task_specs["chief"].instances != 1):
         raise ValueError("exactly one chief task is required")
     for task_type, spec in task_specs.items():
         if spec.instances < 0:
