@@ -109,3 +109,22 @@ def test_run_on_yarn_failure_propagates(tmp_path):
             {"chief": TaskSpec(memory=512, vcores=1)},
             base_dir=str(tmp_path / "app"),
         )
+
+
+@pytest.mark.timeout(180)
+def test_run_on_yarn_pytorch_worker_only(tmp_path):
+    """The reference README's pytorch topology has NO chief
+    (README.md:253-260: workers only, nb_proc_per_worker=2) — the
+    flavor's injected task module must not require one."""
+    model_dir = str(tmp_path / "model")
+    from functools import partial
+    metrics = run_on_yarn(
+        partial(_experiment_fn, model_dir),
+        {
+            "worker": TaskSpec(memory=512, vcores=2, instances=1,
+                               nb_proc_per_worker=2),
+        },
+        base_dir=str(tmp_path / "app"),
+    )
+    assert metrics is not None
+    assert os.path.exists(os.path.join(model_dir, "model_2.pt"))
