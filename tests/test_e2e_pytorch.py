@@ -128,3 +128,37 @@ def test_run_on_yarn_pytorch_worker_only(tmp_path):
     )
     assert metrics is not None
     assert os.path.exists(os.path.join(model_dir, "model_2.pt"))
+
+
+@pytest.mark.timeout(180)
+def test_run_on_yarn_retry_recovers(tmp_path):
+    """nb_retries end to end (not mocked): the task crashes on try 0 and
+    succeeds on try 1 (reference client.py:431-466 whole-app retry)."""
+    flag = str(tmp_path / "first_try_done")
+
+    def experiment_fn():
+        def main_fn(model, loader, device, rank, tb_writer):
+            import os
+            if not os.path.exists(flag):
+                open(flag, "w").close()
+                raise RuntimeError("deliberate first-try failure")
+
+        import torch
+        from torch import nn
+
+        from tf_yarn_amd.pytorch import DataLoaderArgs, PytorchExperiment
+        torch.manual_seed(0)
+        x = torch.randn(16, 4)
+        dataset = torch.utils.data.TensorDataset(x, x.sum(1, keepdim=True))
+        return PytorchExperiment(
+            model=nn.Linear(4, 1), main_fn=main_fn,
+            train_dataset=dataset,
+            dataloader_args=DataLoaderArgs(batch_size=8, pin_memory=False))
+
+    metrics = run_on_yarn(
+        experiment_fn,
+        {"chief": TaskSpec(memory=512, vcores=1)},
+        nb_retries=1,
+        base_dir=str(tmp_path / "app"))
+    assert metrics is not None  # second try succeeded
+    assert os.path.exists(flag)
