@@ -115,3 +115,31 @@ def test_full_topology_with_side_tasks(tmp_path):
     assert os.path.isdir(eval_dir)
     from tf_yarn_amd.estimator.estimator import evaluated_steps
     assert 12 in evaluated_steps(eval_dir)
+
+
+@pytest.mark.timeout(240)
+def test_evaluation_only_run(tmp_path):
+    """The reference README's independent train/eval flow
+    (README.md:351-380): train without an evaluator, then evaluate in a
+    separate evaluator-only run with custom_task_module."""
+    model_dir = str(tmp_path / "model")
+    from functools import partial
+
+    # 1. training-only run (chief alone)
+    metrics = est_run_on_yarn(
+        partial(_experiment_fn, model_dir),
+        {"chief": TaskSpec(memory=512, vcores=1)},
+        base_dir=str(tmp_path / "app1"))
+    assert metrics is not None
+    assert any(f.startswith("model.ckpt-12") or f == "model.ckpt-12"
+               for f in os.listdir(model_dir)), os.listdir(model_dir)
+
+    # 2. evaluation-only run
+    metrics2 = est_run_on_yarn(
+        partial(_experiment_fn, model_dir),
+        {"evaluator": TaskSpec(memory=512, vcores=1)},
+        custom_task_module="tf_yarn_amd.estimator.tasks.evaluator_task",
+        base_dir=str(tmp_path / "app2"))
+    assert metrics2 is not None
+    # evaluator wrote eval events under model_dir/eval
+    assert os.path.isdir(os.path.join(model_dir, "eval"))
