@@ -153,3 +153,65 @@ def test_kv_rendezvous_store_gloo():
                 p.terminate()
         server.stop()
     assert results == {0: 3.0, 1: 3.0}
+
+
+def _worker_no_sync(rank, world_size, kv_addr, out_q):
+    from tf_yarn_amd.parallel import comm
+    from tf_yarn_amd.parallel.ddp import BucketedDataParallel
+    client = KVClient(kv_addr)
+    comm.init_process_group(rank=rank, world_size=world_size,
+                            backend="gloo", kv_client=client)
+    try:
+        model = _make_model()
+        ddp = BucketedDataParallel(model, bucket_cap_mb=1)
+        torch.manual_seed(200 + rank)
+        x1, y1 = torch.randn(8, 16), torch.randn(8, 4)
+        x2, y2 = torch.randn(8, 16), torch.randn(8, 4)
+        with ddp.no_sync():  # accumulation step: NO allreduce
+            nn.functional.mse_loss(ddp(x1), y1).backward()
+        nn.functional.mse_loss(ddp(x2), y2).backward()  # synced step
+        grads = [p.grad.numpy().copy() for p in model.parameters()]
+        out_q.put((rank, grads))
+    finally:
+        comm.destroy_process_group()
+
+
+def test_no_sync_gradient_accumulation():
+    """no_sync parity with torch DDP semantics: grads accumulate locally
+    during no_sync and the next synced backward averages the TOTALS."""
+    world_size = 2
+    server = KVServer()
+    ctx = mp.get_context("spawn")
+    out_q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_no_sync,
+                         args=(r, world_size, server.address, out_q))
+             for r in range(world_size)]
+    for p in procs:
+        p.start()
+    results = {}
+    try:
+        for _ in range(world_size):
+            rank, grads = out_q.get(timeout=120)
+            results[rank] = grads
+    finally:
+        for p in procs:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.terminate()
+        server.stop()
+    # reference: mean over ranks of (g_batch1 + g_batch2)
+    ref_model = _make_model()
+    accum = [torch.zeros_like(p) for p in ref_model.parameters()]
+    for rank in range(world_size):
+        m = _make_model()
+        torch.manual_seed(200 + rank)
+        x1, y1 = torch.randn(8, 16), torch.randn(8, 4)
+        x2, y2 = torch.randn(8, 16), torch.randn(8, 4)
+        nn.functional.mse_loss(m(x1), y1).backward()
+        nn.functional.mse_loss(m(x2), y2).backward()
+        for a, p in zip(accum, m.parameters()):
+            a += p.grad / world_size
+    for rank in range(world_size):
+        for g, ref in zip(results[rank], accum):
+            assert torch.allclose(torch.from_numpy(g), ref, atol=1e-6), \
+                f"rank {rank}: no_sync accumulation diverges"
