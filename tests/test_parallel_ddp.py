@@ -215,3 +215,81 @@ def test_no_sync_gradient_accumulation():
         for g, ref in zip(results[rank], accum):
             assert torch.allclose(torch.from_numpy(g), ref, atol=1e-6), \
                 f"rank {rank}: no_sync accumulation diverges"
+
+
+class _BranchyModel(nn.Module):
+    """`side` is unused in forward when use_side=False."""
+
+    def __init__(self):
+        super().__init__()
+        torch.manual_seed(7)
+        self.main = nn.Linear(16, 4)
+        self.side = nn.Linear(16, 4)
+
+    def forward(self, x, use_side=False):
+        out = self.main(x)
+        if use_side:
+            out = out + self.side(x)
+        return out
+
+
+def _worker_unused(rank, world_size, kv_addr, out_q):
+    from tf_yarn_amd.parallel import comm
+    from tf_yarn_amd.parallel.ddp import BucketedDataParallel
+    client = KVClient(kv_addr)
+    comm.init_process_group(rank=rank, world_size=world_size,
+                            backend="gloo", kv_client=client)
+    try:
+        model = _BranchyModel()
+        ddp = BucketedDataParallel(model, bucket_cap_mb=1,
+                                   find_unused_parameters=True)
+        torch.manual_seed(300 + rank)
+        x, y = torch.randn(8, 16), torch.randn(8, 4)
+        ddp.zero_grad_buffers()
+        nn.functional.mse_loss(ddp(x, use_side=False), y).backward()
+        grads = {n: p.grad.numpy().copy()
+                 for n, p in model.named_parameters()}
+        out_q.put((rank, grads))
+    finally:
+        comm.destroy_process_group()
+
+
+def test_find_unused_parameters_syncs_partial_graph():
+    """A branch unused in forward must not deadlock or error with
+    find_unused_parameters=True; used grads average across ranks and
+    unused grads stay zero (torch DDP semantics)."""
+    world_size = 2
+    server = KVServer()
+    ctx = mp.get_context("spawn")
+    out_q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_unused,
+                         args=(r, world_size, server.address, out_q))
+             for r in range(world_size)]
+    for p in procs:
+        p.start()
+    results = {}
+    try:
+        for _ in range(world_size):
+            rank, grads = out_q.get(timeout=120)
+            results[rank] = grads
+    finally:
+        for p in procs:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.terminate()
+        server.stop()
+    # reference for the used branch
+    ref = {n: torch.zeros_like(p)
+           for n, p in _BranchyModel().named_parameters()}
+    for rank in range(world_size):
+        m = _BranchyModel()
+        torch.manual_seed(300 + rank)
+        x, y = torch.randn(8, 16), torch.randn(8, 4)
+        nn.functional.mse_loss(m(x, use_side=False), y).backward()
+        for n, p in m.named_parameters():
+            if p.grad is not None:
+                ref[n] += p.grad / world_size
+    for rank in range(world_size):
+        for n, g in results[rank].items():
+            assert torch.allclose(torch.from_numpy(g), ref[n],
+                                  atol=1e-6), f"rank {rank} {n}"
