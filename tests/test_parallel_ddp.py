@@ -293,3 +293,50 @@ def test_find_unused_parameters_syncs_partial_graph():
         for n, g in results[rank].items():
             assert torch.allclose(torch.from_numpy(g), ref[n],
                                   atol=1e-6), f"rank {rank} {n}"
+
+
+def _worker_buffers(rank, world_size, kv_addr, out_q):
+    from tf_yarn_amd.parallel import comm
+    from tf_yarn_amd.parallel.ddp import BucketedDataParallel
+    client = KVClient(kv_addr)
+    comm.init_process_group(rank=rank, world_size=world_size,
+                            backend="gloo", kv_client=client)
+    try:
+        torch.manual_seed(50)  # same params everywhere
+        model = nn.Sequential(nn.Linear(8, 8), nn.BatchNorm1d(8))
+        # per-rank different buffer state before wrapping
+        model[1].running_mean.fill_(float(rank + 1))
+        ddp = BucketedDataParallel(model, broadcast_buffers=True)
+        ddp(torch.randn(4, 8))  # forward triggers rank-0 buffer broadcast
+        out_q.put((rank, model[1].running_mean.numpy().copy()))
+    finally:
+        comm.destroy_process_group()
+
+
+def test_broadcast_buffers_syncs_from_rank0():
+    world_size = 2
+    server = KVServer()
+    ctx = mp.get_context("spawn")
+    out_q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_buffers,
+                         args=(r, world_size, server.address, out_q))
+             for r in range(world_size)]
+    for p in procs:
+        p.start()
+    results = {}
+    try:
+        for _ in range(world_size):
+            rank, buf = out_q.get(timeout=120)
+            results[rank] = buf
+    finally:
+        for p in procs:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.terminate()
+        server.stop()
+    # both ranks hold rank 0's PRE-forward buffer value in the broadcast
+    # position... the broadcast happens before forward, so rank 1 sees
+    # rank 0's fill value (1.0) folded into its own post-forward update.
+    assert torch.allclose(torch.from_numpy(results[0]),
+                          torch.from_numpy(results[1]), atol=1e-6), \
+        "buffers diverged across ranks after broadcast_buffers forward"
