@@ -143,3 +143,23 @@ def test_evaluation_only_run(tmp_path):
     assert metrics2 is not None
     # evaluator wrote eval events under model_dir/eval
     assert os.path.isdir(os.path.join(model_dir, "eval"))
+
+
+@pytest.mark.timeout(240)
+def test_ps_strategy_two_shards_cpu(tmp_path):
+    """2 ps shards: exercises the greedy parameter partition and the
+    per-(worker, ps) pair groups with a second shard server."""
+    model_dir = str(tmp_path / "model")
+    from functools import partial
+    metrics = est_run_on_yarn(
+        partial(_experiment_fn, model_dir),
+        {
+            "chief": TaskSpec(memory=512, vcores=1),
+            "ps": TaskSpec(memory=512, vcores=1, instances=2),
+            "worker": TaskSpec(memory=512, vcores=1, instances=2),
+        },
+        base_dir=str(tmp_path / "app"),
+    )
+    assert metrics is not None
+    assert any(n.startswith("model.ckpt-")
+               for n in os.listdir(model_dir))
