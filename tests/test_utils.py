@@ -77,3 +77,30 @@ def test_summary_writer_read_events(tmp_path):
     events = tb.read_events(str(tmp_path))
     tags = {e["tag"] for e in events}
     assert tags == {"m/a", "m/b"}
+
+
+def test_packaging_zip_path(tmp_path):
+    """packaging.zip_path (reference packaging.py:23-37 facade)."""
+    import zipfile
+
+    from tf_yarn_amd import packaging
+    src = tmp_path / "pkg"
+    src.mkdir()
+    (src / "a.py").write_text("x = 1\n")
+    (src / "sub").mkdir()
+    (src / "sub" / "b.py").write_text("y = 2\n")
+    out = packaging.zip_path(str(src), False, str(tmp_path))
+    assert out.endswith(".zip") and __import__("os").path.exists(out)
+    names = zipfile.ZipFile(out).namelist()
+    assert any(n.endswith("a.py") for n in names)
+    assert any(n.endswith("b.py") for n in names)
+
+
+def test_check_env_local_and_remote():
+    """bin/check_env (reference bin/check_hadoop_env.py): local env
+    report + 1-task spawner round trip reporting through the KV store."""
+    from tf_yarn_amd.bin import check_env
+    results = check_env.check_local_env()
+    assert results["gloo_backend"] is True
+    assert "torch" in results
+    assert check_env.launch_remote_check() is True
