@@ -104,3 +104,29 @@ def test_check_env_local_and_remote():
     assert results["gloo_backend"] is True
     assert "torch" in results
     assert check_env.launch_remote_check() is True
+
+
+def test_analyze_rocpd_tool(tmp_path):
+    """scripts/analyze_rocpd.py parses a minimal rocpd schema."""
+    import sqlite3
+    import subprocess
+    import sys
+    db = tmp_path / "r.db"
+    conn = sqlite3.connect(str(db))
+    conn.execute("CREATE TABLE rocpd_info_kernel_symbol "
+                 "(id INTEGER, display_name TEXT)")
+    conn.execute("CREATE TABLE rocpd_kernel_dispatch "
+                 "(kernel_id INTEGER, start INTEGER, end INTEGER)")
+    conn.execute("INSERT INTO rocpd_info_kernel_symbol VALUES (1, 'k1')")
+    conn.executemany("INSERT INTO rocpd_kernel_dispatch VALUES (1, ?, ?)",
+                     [(0, 1000), (2000, 4000)])
+    conn.commit()
+    conn.close()
+    import os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "scripts/analyze_rocpd.py", str(db),
+         "--steps", "2", "--markdown"],
+        cwd=repo, capture_output=True, text=True, timeout=60)
+    assert out.returncode == 0, out.stderr
+    assert "k1" in out.stdout and "us/step" in out.stdout
