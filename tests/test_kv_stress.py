@@ -91,3 +91,53 @@ def test_native_vs_python_throughput():
     print("kv put ops/s:", {k: f"{v:,.0f}" for k, v in results.items()})
     if "native" in results:
         assert results["native"] > 0.5 * results["python"]
+
+
+def test_chaos_mixed_ops_server_stays_consistent():
+    """N threads hammer one server with a random mix of put/get/wait/
+    list/delete/add/cas for ~2 s; the server must stay alive and every
+    ADD must be accounted for exactly once."""
+    import random
+    import threading
+
+    from tf_yarn_amd.kv import KVClient, KVServer
+
+    server = KVServer()
+    n_threads = 8
+    adds_per_thread = 50
+    errors = []
+
+    def worker(tid):
+        try:
+            client = KVClient(server.address)
+            rng = random.Random(tid)
+            for i in range(adds_per_thread):
+                op = rng.randrange(5)
+                key = f"chaos/{rng.randrange(20)}"
+                if op == 0:
+                    client.put(key, f"v{tid}:{i}".encode())
+                elif op == 1:
+                    client.get(key)
+                elif op == 2:
+                    client.list("chaos/")
+                elif op == 3 and rng.random() < 0.3:
+                    client.delete(key)
+                elif op == 4:
+                    client.compare_set(key, b"x", b"y")
+                client.add("chaos_counter", 1)  # always
+        except Exception as e:  # noqa: BLE001
+            errors.append((tid, repr(e)))
+
+    threads = [threading.Thread(target=worker, args=(t,))
+               for t in range(n_threads)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=60)
+    try:
+        assert not errors, errors
+        client = KVClient(server.address)
+        total = client.add("chaos_counter", 0)
+        assert total == n_threads * adds_per_thread, total
+    finally:
+        server.stop()
