@@ -65,8 +65,9 @@ def test_worker_task_passes_iterable_through():
     from tf_yarn_amd.pytorch.experiment import DataLoaderArgs
     from tf_yarn_amd.pytorch.tasks.worker import _create_dataloader
     ds = ShardedIterableDataset([0, 1], _read)
-    loader = _create_dataloader(ds, DataLoaderArgs(batch_size=5),
-                                rank=0, world_size=1)
+    loader = _create_dataloader(
+        ds, DataLoaderArgs(batch_size=5, pin_memory=False),
+        rank=0, world_size=1)
     batches = list(loader)
     assert all(len(b) == 5 for b in batches)
     assert int(torch.cat(batches).max()) == 19
