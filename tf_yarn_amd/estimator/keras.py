@@ -134,10 +134,11 @@ class KerasModel:
                 loss = self.loss_fn(out, by)
                 loss.backward()
                 self.optimizer.step()
-                epoch_loss += float(loss)
+                loss_val = float(loss.detach())
+                epoch_loss += loss_val
                 n_b += 1
                 for cb in callbacks:
-                    cb.on_batch_end(b, {"loss": float(loss)}, self)
+                    cb.on_batch_end(b, {"loss": loss_val}, self)
                 if steps_per_epoch is not None and n_b >= steps_per_epoch:
                     break
             logs = {"loss": epoch_loss / max(1, n_b)}
