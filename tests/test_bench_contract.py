@@ -4,10 +4,17 @@ the required fields, both single-process and under torch.distributed.run
 
 import json
 import os
+import socket
 import subprocess
 import sys
 
 import pytest
+
+
+def _free_port() -> str:
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return str(s.getsockname()[1])
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
@@ -51,7 +58,7 @@ def test_bench_under_torchrun_2_ranks():
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29517", "bench.py", "--gpus", "2",
+         "--master-port", _free_port(), "bench.py", "--gpus", "2",
          "--steps", "2", "--warmup", "1", "--batch", "64",
          "--table-rows", "1000", "--dtype", "fp32"],
         cwd=REPO, env=env, capture_output=True, text=True, timeout=280)
@@ -66,7 +73,7 @@ def test_bench_under_torchrun_4_ranks_uneven_shards():
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
-         "--master-port", "29547", "bench.py", "--gpus", "4",
+         "--master-port", _free_port(), "bench.py", "--gpus", "4",
          "--steps", "2", "--warmup", "1", "--batch", "32",
          "--table-rows", "500", "--dtype", "fp32"],
         cwd=REPO, capture_output=True, text=True, timeout=280)
@@ -81,7 +88,7 @@ def test_bench_under_torchrun_8_ranks():
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
-         "--master-port", "29587", "bench.py", "--gpus", "8",
+         "--master-port", _free_port(), "bench.py", "--gpus", "8",
          "--steps", "2", "--warmup", "1", "--batch", "16",
          "--table-rows", "200", "--dtype", "fp32"],
         cwd=REPO, capture_output=True, text=True, timeout=280)
