@@ -40,3 +40,14 @@ def test_reference_pytorch_exports():
                  "RunFailed", "Metrics", "TaskSpec", "NodeLabel",
                  "get_safe_experiment_fn", "DistributedDataParallelArgs"):
         assert hasattr(t, name), name
+
+
+def test_tensorflow_alias_submodules_importable():
+    """custom_task_module strings through the alias package must resolve
+    (e.g. "tf_yarn_amd.tensorflow.tasks.evaluator_task")."""
+    import importlib
+    for name in ("tf_yarn_amd.tensorflow.client",
+                 "tf_yarn_amd.tensorflow.cluster",
+                 "tf_yarn_amd.tensorflow.metrics",
+                 "tf_yarn_amd.tensorflow.tasks.evaluator_task"):
+        assert importlib.import_module(name) is not None, name
