@@ -162,3 +162,50 @@ def test_run_on_yarn_retry_recovers(tmp_path):
         base_dir=str(tmp_path / "app"))
     assert metrics is not None  # second try succeeded
     assert os.path.exists(flag)
+
+
+@pytest.mark.timeout(180)
+def test_run_on_yarn_with_parquet_dataset(tmp_path):
+    """PytorchExperiment over a ParquetDataset: the iterable dataset
+    shards itself by rank inside the spawned workers (reference
+    worker.py:54-65 + parquet_dataset.py integration)."""
+    pq = pytest.importorskip("pyarrow.parquet")
+    import pyarrow as pa
+
+    pfile = tmp_path / "data.parquet"
+    table = pa.table({"x": [float(i) for i in range(160)]})
+    pq.write_table(table, str(pfile))
+    out_dir = tmp_path / "out"
+    out_dir.mkdir()
+
+    def experiment_fn():
+        import torch
+        from torch import nn
+
+        from tf_yarn_amd.pytorch import DataLoaderArgs, PytorchExperiment
+        from tf_yarn_amd.pytorch.parquet_dataset import ParquetDataset
+
+        def main_fn(model, loader, device, rank, tb_writer):
+            rows = []
+            for batch in loader:
+                rows += [float(v) for v in batch.column("x").to_pylist()]
+            (out_dir / f"rank{rank}.txt").write_text(
+                ",".join(str(int(r)) for r in rows))
+
+        ds = ParquetDataset([str(pfile)], batch_size=10)
+        return PytorchExperiment(
+            model=nn.Linear(2, 1), main_fn=main_fn, train_dataset=ds,
+            dataloader_args=DataLoaderArgs(batch_size=None,
+                                           pin_memory=False))
+
+    metrics = run_on_yarn(
+        experiment_fn,
+        {"worker": TaskSpec(memory=512, vcores=1, instances=2)},
+        base_dir=str(tmp_path / "app"))
+    assert metrics is not None
+    seen = []
+    for rank in range(2):
+        vals = (out_dir / f"rank{rank}.txt").read_text().split(",")
+        assert len(vals) == 80, f"rank {rank} got {len(vals)} rows"
+        seen.append(set(vals))
+    assert seen[0].isdisjoint(seen[1])

@@ -42,6 +42,14 @@ class ParquetDataset(torch.utils.data.IterableDataset):
             self.world_size = 1
 
     def __iter__(self):
+        # Defensive: a dataset constructed before init_process_group
+        # (default rank 0 / world 1) picks up the live group at first
+        # iteration, so sharding is correct even when the experiment was
+        # materialized early.
+        if self.world_size == 1 and dist.is_available() \
+                and dist.is_initialized():
+            self.rank = dist.get_rank()
+            self.world_size = dist.get_world_size()
         worker_info = torch.utils.data.get_worker_info()
         n_loaders = worker_info.num_workers if worker_info else 1
         loader_id = worker_info.id if worker_info else 0

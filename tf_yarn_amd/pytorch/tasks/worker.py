@@ -70,6 +70,10 @@ def _create_dataloader(dataset,
         kwargs.pop("prefetch_factor")
         if kwargs.get("num_workers", 0) == 0:
             kwargs["pin_memory"] = kwargs.get("pin_memory", False)
+    if kwargs.get("batch_size") is None:
+        # pre-batched dataset (e.g. ParquetDataset yields whole batches):
+        # auto-batching off; drop_last/ sampler options don't apply
+        kwargs.pop("drop_last", None)
     if isinstance(dataset, torch.utils.data.IterableDataset):
         return torch.utils.data.DataLoader(dataset, **kwargs)
     sampler = torch.utils.data.distributed.DistributedSampler(
@@ -88,8 +92,7 @@ def _upload_tensorboard_logs(local_dir: str, dest_dir: str,
         logger.exception("failed to upload tensorboard logs")
 
 
-def _train(experiment: PytorchExperiment,
-           client: KVClient,
+def _train(client: KVClient,
            device: str,
            rank: int,
            world_size: int) -> None:
@@ -102,6 +105,11 @@ def _train(experiment: PytorchExperiment,
                             backend=backend, device=device,
                             kv_client=client)
     try:
+        # Materialize the experiment AFTER the process group is up, like
+        # the reference (worker.py:101-103): datasets that detect
+        # rank/world at construction (ParquetDataset) see the live group.
+        experiment = _task_commons._get_experiment(client)
+        assert isinstance(experiment, PytorchExperiment), type(experiment)
         model = experiment.model.to(device)
         ddp_args = experiment.ddp_args or DistributedDataParallelArgs()
         ddp_model = BucketedDataParallel(
@@ -172,19 +180,17 @@ def main() -> None:
 
 def _run_single(client: KVClient, rank: int, world_size: int,
                 gpu_ids: List[int]) -> None:
-    experiment = _task_commons._get_experiment(client)
     device = _get_device(gpu_ids, 0)
-    _train(experiment, client, device, rank, world_size)
+    _train(client, device, rank, world_size)
 
 
 def _spawned_entry(local_rank: int, kv_addr: str, rank_base: int,
                    world_size: int, gpu_ids: List[int]) -> None:
     _task_commons.setup_logging()
     client = KVClient(kv_addr)
-    experiment = _task_commons._get_experiment(client)
     rank = rank_base + local_rank
     device = _get_device(gpu_ids, local_rank)
-    _train(experiment, client, device, rank, world_size)
+    _train(client, device, rank, world_size)
 
 
 if __name__ == "__main__":
