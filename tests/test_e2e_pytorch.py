@@ -209,3 +209,40 @@ def test_run_on_yarn_with_parquet_dataset(tmp_path):
         assert len(vals) == 80, f"rank {rank} got {len(vals)} rows"
         seen.append(set(vals))
     assert seen[0].isdisjoint(seen[1])
+
+
+@pytest.mark.timeout(180)
+def test_run_on_yarn_with_sharded_iterable_dataset(tmp_path):
+    """ShardedIterableDataset (the webdataset-branch analog) through the
+    spawned flavor: two workers stream disjoint shards."""
+    out_dir = tmp_path / "out"
+    out_dir.mkdir()
+
+    def experiment_fn():
+        from torch import nn
+
+        from tf_yarn_amd.pytorch import DataLoaderArgs, PytorchExperiment
+        from tf_yarn_amd.pytorch.web_dataset import ShardedIterableDataset
+
+        def read_shard(shard):
+            return range(shard * 10, shard * 10 + 10)
+
+        def main_fn(model, loader, device, rank, tb_writer):
+            seen = [int(v) for batch in loader for v in batch]
+            (out_dir / f"rank{rank}.txt").write_text(
+                ",".join(map(str, sorted(seen))))
+
+        ds = ShardedIterableDataset(list(range(4)), read_shard)
+        return PytorchExperiment(
+            model=nn.Linear(2, 1), main_fn=main_fn, train_dataset=ds,
+            dataloader_args=DataLoaderArgs(batch_size=5,
+                                           pin_memory=False))
+
+    metrics = run_on_yarn(
+        experiment_fn,
+        {"worker": TaskSpec(memory=512, vcores=1, instances=2)},
+        base_dir=str(tmp_path / "app"))
+    assert metrics is not None
+    r0 = set((out_dir / "rank0.txt").read_text().split(","))
+    r1 = set((out_dir / "rank1.txt").read_text().split(","))
+    assert len(r0) == 20 and len(r1) == 20 and r0.isdisjoint(r1)
