@@ -246,3 +246,50 @@ def test_run_on_yarn_with_sharded_iterable_dataset(tmp_path):
     r0 = set((out_dir / "rank0.txt").read_text().split(","))
     r1 = set((out_dir / "rank1.txt").read_text().split(","))
     assert len(r0) == 20 and len(r1) == 20 and r0.isdisjoint(r1)
+
+
+@pytest.mark.timeout(180)
+def test_checkpoint_resume_across_runs(tmp_path):
+    """The reference resume pattern (pytorch_distributed_example.py:82-98):
+    run 2 resumes from the latest model_{epoch}.pt of run 1."""
+    model_dir = str(tmp_path / "ckpts")
+    out = tmp_path / "resumed_epoch.txt"
+
+    def experiment_fn():
+        import torch
+        from torch import nn
+
+        from tf_yarn_amd.pytorch import DataLoaderArgs, PytorchExperiment
+        from tf_yarn_amd.pytorch import model_ckpt
+
+        def main_fn(model, loader, device, rank, tb_writer):
+            opt = torch.optim.SGD(model.parameters(), lr=0.05)
+            ckpt = model_ckpt.load_latest_ckpt(model_dir, model, opt,
+                                               device)
+            start_epoch = (ckpt["epoch"] + 1) if ckpt else 0
+            for epoch in range(start_epoch, start_epoch + 2):
+                for x, y in loader:
+                    opt.zero_grad()
+                    nn.functional.mse_loss(model(x), y).backward()
+                    opt.step()
+            if rank == 0:
+                model_ckpt.save_ckpt(model_dir, model, opt,
+                                     epoch=start_epoch + 1)
+                out.write_text(str(start_epoch))
+
+        torch.manual_seed(0)
+        x = torch.randn(32, 4)
+        dataset = torch.utils.data.TensorDataset(x, x.sum(1, keepdim=True))
+        return PytorchExperiment(
+            model=nn.Linear(4, 1), main_fn=main_fn, train_dataset=dataset,
+            dataloader_args=DataLoaderArgs(batch_size=8, pin_memory=False))
+
+    for i in range(2):
+        metrics = run_on_yarn(
+            experiment_fn,
+            {"chief": TaskSpec(memory=512, vcores=1)},
+            base_dir=str(tmp_path / f"app{i}"))
+        assert metrics is not None
+    # run 1 started at 0, saved model_1.pt; run 2 resumed at epoch 2
+    assert out.read_text() == "2"
+    assert os.path.exists(os.path.join(model_dir, "model_3.pt"))
