@@ -53,3 +53,18 @@ def test_check_env_round_trip(tmp_path):
     assert "torch" in local
     assert local["gloo_backend"] is True
     assert launch_remote_check(str(tmp_path / "app")) is True
+
+
+@pytest.mark.timeout(180)
+def test_distributed_flavor_multiproc_per_worker(tmp_path):
+    """nb_proc_per_worker=2: rank math task_id * nb_proc + local_rank
+    (reference distributed/task.py:37-55) across spawned sub-processes."""
+    metrics = run_on_yarn(
+        _train_fn,
+        {
+            "worker": TaskSpec(memory=512, vcores=2, instances=2,
+                               nb_proc_per_worker=2),
+        },
+        base_dir=str(tmp_path / "app"),
+    )
+    assert metrics is not None  # 4-rank allreduce inside asserted 1+2+3+4
