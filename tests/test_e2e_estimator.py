@@ -163,3 +163,60 @@ def test_ps_strategy_two_shards_cpu(tmp_path):
     assert metrics is not None
     assert any(n.startswith("model.ckpt-")
                for n in os.listdir(model_dir))
+
+
+def _keras_experiment_fn(model_dir):
+    import os
+
+    import torch
+    from torch import nn
+
+    os.makedirs(model_dir, exist_ok=True)
+
+    from tf_yarn_amd.estimator.keras import KerasModel, ModelCheckpoint
+    from tf_yarn_amd.estimator.keras_experiment import KerasExperiment
+
+    torch.manual_seed(0)
+    model = KerasModel(nn.Sequential(nn.Linear(4, 8), nn.ReLU(),
+                                     nn.Linear(8, 1)))
+    # Adadelta: the reference README's Keras optimizer (README.md:106)
+    model.compile(optimizer="adadelta", loss="mse")
+
+    def input_data_fn():
+        torch.manual_seed(1)
+        return torch.randn(128, 4)
+
+    def target_data_fn():
+        torch.manual_seed(1)
+        return torch.randn(128, 4).sum(dim=1, keepdim=True)
+
+    return KerasExperiment(
+        model=model, model_dir=model_dir,
+        train_params={"epochs": 2, "batch_size": 16,
+                      "callbacks": [ModelCheckpoint(
+                          os.path.join(model_dir, "checkpoint-{epoch}"))]},
+        input_data_fn=input_data_fn,
+        target_data_fn=target_data_fn,
+        validation_data_fn=None)
+
+
+@pytest.mark.timeout(240)
+def test_keras_allreduce_cpu(tmp_path):
+    """KerasExperiment through the allreduce task module — the
+    reference's native_keras_with_gloo_example flow (chief doubles as
+    rendezvous driver, only the chief keeps ModelCheckpoint)."""
+    model_dir = str(tmp_path / "model")
+    from functools import partial
+    metrics = est_run_on_yarn(
+        partial(_keras_experiment_fn, model_dir),
+        {
+            "chief": TaskSpec(memory=512, vcores=1),
+            "worker": TaskSpec(memory=512, vcores=1, instances=1),
+        },
+        custom_task_module="tf_yarn_amd.estimator.tasks.allred_task",
+        base_dir=str(tmp_path / "app"),
+    )
+    assert metrics is not None
+    # only the chief wrote the Keras checkpoint-{epoch} layout
+    names = os.listdir(model_dir)
+    assert any(n.startswith("checkpoint-") for n in names), names
