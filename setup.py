@@ -50,6 +50,8 @@ setup(
     description=("MI355X-native distributed-training launcher with "
                  "criteo/tf-yarn's capabilities"),
     packages=find_packages(include=["tf_yarn_amd", "tf_yarn_amd.*"]),
+    package_data={"tf_yarn_amd": ["default.log.conf"]},
+    python_requires=">=3.9",
     ext_modules=[ext, kv_ext],
     cmdclass={"build_ext": BuildExtension},
 )
