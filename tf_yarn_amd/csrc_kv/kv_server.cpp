@@ -209,6 +209,7 @@ class KvServer {
     while (running_) {
       uint32_t frame;
       if (!read_exact(fd, &frame, 4)) break;
+      if (frame > (256u << 20)) break;  // corrupt length guard
       std::vector<uint8_t> body(frame);
       if (!read_exact(fd, body.data(), frame)) break;
       if (frame < 3) break;
@@ -219,6 +220,7 @@ class KvServer {
       std::string key(reinterpret_cast<char*>(body.data() + 3), klen);
       uint64_t plen;
       memcpy(&plen, body.data() + 3 + klen, 8);
+      if (3 + static_cast<uint64_t>(klen) + 8 + plen > frame) break;
       std::string payload(
           reinterpret_cast<char*>(body.data() + 3 + klen + 8), plen);
 

@@ -28,6 +28,7 @@ emulation backs the CPU tests (`_all_to_all_single`).
 from __future__ import annotations
 
 import logging
+import weakref
 from typing import List, Optional, Tuple
 
 import torch
@@ -45,7 +46,9 @@ def _world_rank(group) -> Tuple[int, int]:
     return 1, 0
 
 
-_ALLTOALL_MODE: dict = {}  # per-group: "native" | "emulate"
+# Per-group probe cache, weak-keyed on the ProcessGroup object itself so a
+# destroyed+gc'd group can never alias a new group's entry (id() reuse).
+_ALLTOALL_MODE: "weakref.WeakKeyDictionary" = weakref.WeakKeyDictionary()
 
 
 def _alltoall_self_check(group=None) -> str:
@@ -88,10 +91,13 @@ def _all_to_all_single(output: torch.Tensor, input_: torch.Tensor,
     one-time native self-check on other backends."""
     backend = dist.get_backend(group)
     if backend != "gloo":
-        key = id(group)
-        if key not in _ALLTOALL_MODE:
-            _ALLTOALL_MODE[key] = _alltoall_self_check(group)
-        if _ALLTOALL_MODE[key] == "native":
+        key = (group if group is not None
+               else dist.distributed_c10d._get_default_group())
+        mode = _ALLTOALL_MODE.get(key)
+        if mode is None:
+            mode = _alltoall_self_check(group)
+            _ALLTOALL_MODE[key] = mode
+        if mode == "native":
             dist.all_to_all_single(output, input_,
                                    output_split_sizes=out_splits,
                                    input_split_sizes=in_splits,

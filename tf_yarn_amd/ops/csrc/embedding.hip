@@ -21,6 +21,7 @@
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 #include "common.h"
+#include <cstdlib>
 
 namespace {
 
@@ -215,6 +216,26 @@ __global__ void emb_bwd_sgd_sorted_kernel(
   }
 }
 
+// Out-of-range ids silently corrupt adjacent table rows via atomicAdd;
+// MIYARN_DEBUG_BOUNDS=1 turns on a host-side min/max validation (two
+// small reductions + sync per call — debug only, off in production).
+bool bounds_debug_enabled() {
+  static const bool on = [] {
+    const char* v = std::getenv("MIYARN_DEBUG_BOUNDS");
+    return v != nullptr && v[0] != '\0' && v[0] != '0';
+  }();
+  return on;
+}
+
+void debug_check_ids(const torch::Tensor& ids, int64_t rows) {
+  if (!bounds_debug_enabled() || ids.numel() == 0) return;
+  const int64_t mn = ids.min().item<int64_t>();
+  const int64_t mx = ids.max().item<int64_t>();
+  TORCH_CHECK(mn >= 0 && mx < rows, "embedding ids out of range: [", mn,
+              ", ", mx, "] vs table rows ", rows,
+              " (MIYARN_DEBUG_BOUNDS check)");
+}
+
 void check_emb(const torch::Tensor& table, const torch::Tensor& ids,
                int64_t dim) {
   TORCH_CHECK(table.is_cuda() && table.is_contiguous() &&
@@ -232,6 +253,7 @@ torch::Tensor emb_fwd(torch::Tensor table, torch::Tensor ids,
   const int64_t dim = table.size(1);
   const int64_t n = ids.numel();
   check_emb(table, ids, dim);
+  debug_check_ids(ids, table.size(0));
   auto out = torch::empty(
       {n, dim}, table.options().dtype(
           out_bf16 ? torch::kBFloat16 : torch::kFloat32));
@@ -270,6 +292,7 @@ void emb_bwd_sgd(torch::Tensor table, torch::Tensor ids, torch::Tensor grad,
   const int64_t dim = table.size(1);
   const int64_t n = ids.numel();
   check_emb(table, ids, dim);
+  debug_check_ids(ids, table.size(0));
   TORCH_CHECK(grad.is_cuda() && grad.is_contiguous() &&
               grad.numel() == n * dim, "grad shape mismatch");
   auto stream = c10::hip::getCurrentHIPStream().stream();
@@ -311,6 +334,7 @@ void emb_bwd_dense(torch::Tensor grad_table, torch::Tensor ids,
   const int64_t dim = grad_table.size(1);
   const int64_t n = ids.numel();
   check_emb(grad_table, ids, dim);
+  debug_check_ids(ids, grad_table.size(0));
   TORCH_CHECK(grad.is_cuda() && grad.is_contiguous() &&
               grad.numel() == n * dim, "grad shape mismatch");
   auto stream = c10::hip::getCurrentHIPStream().stream();
@@ -356,6 +380,7 @@ torch::Tensor emb_gather_sum(torch::Tensor table, torch::Tensor ids,
   TORCH_CHECK(ids.is_cuda() && ids.is_contiguous() &&
               ids.scalar_type() == torch::kInt64, "ids must be int64 GPU");
   TORCH_CHECK(ids.numel() % batch == 0, "ids size not divisible by batch");
+  debug_check_ids(ids, table.numel());
   const int F = static_cast<int>(ids.numel() / batch);
   auto out = torch::empty({batch}, table.options().dtype(
       out_bf16 ? torch::kBFloat16 : torch::kFloat32));
@@ -385,6 +410,7 @@ void emb_scatter_sum(torch::Tensor table, torch::Tensor ids,
               ids.scalar_type() == torch::kInt64, "ids must be int64 GPU");
   const int64_t batch = grad.numel();
   TORCH_CHECK(ids.numel() % batch == 0, "ids size not divisible by batch");
+  debug_check_ids(ids, table.numel());
   const int F = static_cast<int>(ids.numel() / batch);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   int grid = miyarn_grid(batch * F);
@@ -409,6 +435,7 @@ void emb_fwd_into(torch::Tensor table, torch::Tensor ids,
   const int64_t dim = table.size(1);
   const int64_t n = ids.numel();
   check_emb(table, ids, dim);
+  debug_check_ids(ids, table.size(0));
   TORCH_CHECK(out.is_cuda() && out.is_contiguous() && out.dim() == 2,
               "out must be a contiguous 2D GPU tensor");
   TORCH_CHECK(dim % 4 == 0 && col_offset % 4 == 0 &&
@@ -443,6 +470,7 @@ void emb_bwd_sgd_sorted(torch::Tensor table, torch::Tensor sorted_ids,
   const int64_t dim = table.size(1);
   const int64_t n = sorted_ids.numel();
   check_emb(table, sorted_ids, dim);
+  debug_check_ids(sorted_ids, table.size(0));
   TORCH_CHECK(dim % 4 == 0, "sorted scatter needs dim % 4 == 0");
   TORCH_CHECK(grad.is_cuda() && grad.is_contiguous() &&
               grad.numel() == n * dim, "grad shape mismatch");

@@ -8,6 +8,7 @@ from __future__ import annotations
 
 import logging
 import os
+import pickle
 import re
 import tempfile
 from typing import Any, Dict, Optional
@@ -46,11 +47,23 @@ def load_latest_ckpt(model_dir: str, model, optimizer=None,
 
 
 def load_ckpt(ckpt_path: str, model, optimizer=None,
-              device: str = "cpu") -> Dict[str, Any]:
-    """Reference ``model_ckpt.py:42-52``."""
+              device: str = "cpu", weights_only: bool = True) -> Dict[str, Any]:
+    """Reference ``model_ckpt.py:42-52``.
+
+    ``weights_only=True`` by default (safe unpickling); pass ``False``
+    only for trusted checkpoints carrying arbitrary non-tensor extras.
+    """
     fs, path = resolve_filesystem_and_path(ckpt_path)
     with fs.open(path, "rb") as fd:
-        state = torch.load(fd, map_location=device, weights_only=False)
+        try:
+            state = torch.load(fd, map_location=device,
+                               weights_only=weights_only)
+        except pickle.UnpicklingError:
+            if weights_only:
+                raise RuntimeError(
+                    f"{ckpt_path} contains non-tensor objects; reload with "
+                    "weights_only=False if you trust its source") from None
+            raise
     _unwrap_model(model).load_state_dict(state["model"])
     if optimizer is not None and "optimizer" in state:
         optimizer.load_state_dict(state["optimizer"])
