@@ -47,7 +47,7 @@ FIRST_BUCKET_MB = 4
 
 class _Bucket:
     __slots__ = ("params", "buffer", "views", "pending", "work", "index",
-                 "offsets")
+                 "offsets", "seen")
 
     def __init__(self, index: int, params: List[nn.Parameter]):
         self.index = index
@@ -64,6 +64,11 @@ class _Bucket:
                 self.buffer[offset:offset + p.numel()].view_as(p))
             offset += p.numel()
         self.pending = len(params)
+        # Per-slot readiness: which params actually produced a grad this
+        # backward.  Needed to (a) name unused params in the
+        # find_unused_parameters=False error, and (b) zero only the
+        # never-written slots when find_unused_parameters=True.
+        self.seen = [False] * len(params)
         self.work: Optional[dist.Work] = None
 
 
@@ -102,6 +107,7 @@ class BucketedDataParallel(nn.Module):
         # a bucket whose grads complete early waits for its predecessors.
         self._ready: List[bool] = []
         self._next_launch = 0
+        self._param_names = {p: n for n, p in module.named_parameters()}
         if self._world_size > 1:
             self._build_buckets()
             self._ready = [False] * len(self._buckets)
@@ -194,7 +200,9 @@ class BucketedDataParallel(nn.Module):
             # view and re-link so later iterations accumulate in place.
             bucket.views[slot].copy_(param.grad)
             param.grad = bucket.views[slot]
-        bucket.pending -= 1
+        if not bucket.seen[slot]:
+            bucket.seen[slot] = True
+            bucket.pending -= 1
         if bucket.pending == 0:
             self._ready[bucket.index] = True
             self._launch_ready_in_order()
@@ -221,17 +229,32 @@ class BucketedDataParallel(nn.Module):
 
     def _finalize_backward(self) -> None:
         self._backward_queued = False
-        for bucket in self._buckets:
-            if bucket.pending != 0 \
-                    and bucket.pending == len(bucket.params) \
-                    and not self.find_unused_parameters:
-                bucket.pending = len(bucket.params)
+        if not self.find_unused_parameters:
+            unused = [self._param_names.get(p, f"<param {slot}>")
+                      for bucket in self._buckets if bucket.pending
+                      for slot, p in enumerate(bucket.params)
+                      if not bucket.seen[slot]]
+            if unused:
+                # A partial or empty bucket would allreduce stale/zero view
+                # contents — silent wrong gradients.  Error loudly (torch
+                # DDP behavior) but FIRST restore reducer invariants: drain
+                # every in-flight work and reset per-step state so the
+                # collective launch order stays rank-consistent if the
+                # caller survives the exception.
+                self._reset_step_state()
                 raise RuntimeError(
-                    f"bucket {bucket.index} has no ready grads; "
-                    "pass find_unused_parameters=True if the model "
-                    "has parameters unused in forward")
-        # launch every not-yet-launched bucket (unused / partial grads
-        # keep their zero or stale-view contents), still in index order
+                    "some parameters received no gradient this backward "
+                    f"(unused in forward?): {unused[:16]}"
+                    f"{' ...' if len(unused) > 16 else ''}; pass "
+                    "find_unused_parameters=True to tolerate this")
+        else:
+            # Zero never-written slots in unlaunched buckets so they
+            # contribute exactly zero (not stale content) to the allreduce.
+            for bucket in self._buckets[self._next_launch:]:
+                for slot in range(len(bucket.params)):
+                    if not bucket.seen[slot]:
+                        bucket.views[slot].zero_()
+        # launch every not-yet-launched bucket, still in index order
         while self._next_launch < len(self._buckets):
             self._launch_bucket(self._buckets[self._next_launch])
             self._next_launch += 1
@@ -244,6 +267,18 @@ class BucketedDataParallel(nn.Module):
             if scale_needed:
                 bucket.buffer.div_(self._world_size)
             bucket.pending = len(bucket.params)
+            bucket.seen = [False] * len(bucket.params)
+        self._ready = [False] * len(self._buckets)
+        self._next_launch = 0
+
+    def _reset_step_state(self) -> None:
+        """Drain in-flight works and reset all per-step reducer state."""
+        for bucket in self._buckets:
+            if bucket.work is not None:
+                bucket.work.wait()
+                bucket.work = None
+            bucket.pending = len(bucket.params)
+            bucket.seen = [False] * len(bucket.params)
         self._ready = [False] * len(self._buckets)
         self._next_launch = 0
 
