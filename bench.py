@@ -60,6 +60,52 @@ def log(msg: str) -> None:
     print(f"[bench] {msg}", file=sys.stderr, flush=True)
 
 
+def _ensure_master_port(rank: int) -> None:
+    """Pick MASTER_PORT when the launcher did not set one (torchrun always
+    does).  A fixed fallback (29500) turns a stale process on the box into
+    a rendezvous hang; instead rank 0 probes for a free port starting from
+    a base and publishes it through a file all local ranks share (the
+    bench contract is single-node)."""
+    if "MASTER_PORT" in os.environ:
+        return
+    import socket
+    import tempfile
+    rdv = os.path.join(tempfile.gettempdir(),
+                       f"miyarn_bench_port_{os.environ.get('MIYARN_RDV_TAG', 'default')}")
+    if rank == 0:
+        port = None
+        for cand in range(29500, 29600):
+            with socket.socket(socket.AF_INET, socket.SOCK_STREAM) as s:
+                try:
+                    s.bind(("127.0.0.1", cand))
+                    port = cand
+                    break
+                except OSError:
+                    continue
+        if port is None:
+            raise RuntimeError("no free port in 29500-29599")
+        with open(rdv + ".tmp", "w") as f:
+            f.write(str(port))
+        os.replace(rdv + ".tmp", rdv)
+        os.environ["MASTER_PORT"] = str(port)
+        log(f"rank 0 chose MASTER_PORT={port}")
+    else:
+        start = time.time()
+        deadline = start + 120
+        while time.time() < deadline:
+            try:
+                # ignore stale files from earlier runs: rank 0 writes its
+                # file AFTER every rank has started
+                if os.path.getmtime(rdv) >= start - 5:
+                    with open(rdv) as f:
+                        os.environ["MASTER_PORT"] = f.read().strip()
+                    return
+            except (FileNotFoundError, ValueError, OSError):
+                pass
+            time.sleep(0.2)
+        raise RuntimeError("timed out waiting for rank 0's MASTER_PORT file")
+
+
 def main() -> None:
     parser = argparse.ArgumentParser()
     parser.add_argument("--gpus", type=int, default=1)
@@ -70,6 +116,9 @@ def main() -> None:
     parser.add_argument("--table-rows", type=int,
                         default=TABLE_ROWS_PER_FEATURE,
                         help="rows per categorical table (tests use small)")
+    parser.add_argument("--probe-steps", type=int, default=3,
+                        help="instrumented steps AFTER the timed region "
+                             "for the comm_breakdown JSON field (0=off)")
     parser.add_argument("--graphs", action="store_true",
                         help="capture the step in a hipGraph (measured "
                              "2.39 vs 2.10 ms/step eager on MI355X — "
@@ -89,7 +138,7 @@ def main() -> None:
 
     if world_size > 1:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        os.environ.setdefault("MASTER_PORT", "29500")
+        _ensure_master_port(rank)
         dist.init_process_group(backend, rank=rank, world_size=world_size)
 
     compute_dtype = (torch.bfloat16 if args.dtype == "bf16" and use_gpu
@@ -197,6 +246,44 @@ def main() -> None:
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
     elapsed = float(t.item())
 
+    # -- per-phase comm/compute breakdown (extra instrumented steps AFTER
+    # the timed region; headline number is unperturbed).  Makes a poor
+    # scaling curve attributable: alltoall ids/vec/wide/grad, wide
+    # allgather, dense-reducer drain, plus fwd/bwd/opt phase times.
+    breakdown = None
+    if args.probe_steps > 0:
+        from tf_yarn_amd.utils import commprobe
+        commprobe.reset()
+        commprobe.enable()
+        phases = {"forward": 0.0, "backward": 0.0, "opt_sparse": 0.0}
+        for i in range(args.probe_steps):
+            dense, ids, labels = batches[i % N_DATA_BATCHES]
+            opt.zero_grad(set_to_none=set_to_none)
+            with commprobe.span("phase_forward"):
+                loss = ddp(dense, ids, labels=labels)
+            with commprobe.span("phase_backward"):
+                loss.backward()
+            with commprobe.span("phase_opt_sparse"):
+                module.start_sparse_sync()
+                opt.step()
+                module.finish_sparse_sync(lr)
+        del phases
+        summ = commprobe.summary()
+        commprobe.disable()
+        n = float(args.probe_steps)
+        breakdown = {tag: {"ms_per_step": v["ms"] / n,
+                           "calls_per_step": v["count"] / n}
+                     for tag, v in sorted(summ.items())}
+        # cross-rank MAX of total a2a time: a slow link shows up here
+        a2a_ms = sum(v["ms"] for tag, v in summ.items()
+                     if tag.startswith(("a2a_", "ag_"))) / n
+        tt = torch.tensor([a2a_ms], dtype=torch.float64,
+                          device=device if backend == "nccl" else "cpu")
+        if world_size > 1:
+            dist.all_reduce(tt, op=dist.ReduceOp.MAX)
+        breakdown["sparse_comm_max_over_ranks"] = {
+            "ms_per_step": float(tt.item()), "calls_per_step": 1.0}
+
     if rank == 0:
         ms_per_step = elapsed / args.steps * 1000.0
         examples_per_sec = args.batch * world_size * args.steps / elapsed
@@ -226,6 +313,7 @@ def main() -> None:
                 "final_loss": float(last_loss.item())
                 if last_loss is not None else None,
             },
+            "comm_breakdown": breakdown,
         }
         print(json.dumps(result), flush=True)
 

@@ -36,6 +36,7 @@ import torch.distributed as dist
 from torch import nn
 
 from tf_yarn_amd import ops
+from tf_yarn_amd.utils import commprobe
 
 logger = logging.getLogger(__name__)
 
@@ -157,8 +158,9 @@ class _ShardedLookup(torch.autograd.Function):
         in_splits = [B * len(m.feats_of[s]) for s in range(W)]
         n_own = B * m.f_own
         ids_recv = torch.empty(W * n_own, dtype=torch.int64, device=device)
-        _all_to_all_single(ids_recv, ids_send, [n_own] * W, in_splits,
-                           m.group)
+        with commprobe.span("a2a_ids"):
+            _all_to_all_single(ids_recv, ids_send, [n_own] * W, in_splits,
+                               m.group)
         # local row offsets: layout [peer][b][j], j = feature slot
         flat_local = ids_recv + m.own_offsets_tiled[:ids_recv.numel()]
 
@@ -169,9 +171,11 @@ class _ShardedLookup(torch.autograd.Function):
 
         # ---- route vectors back -----------------------------------------
         vec_recv = torch.empty(B * m.F * D, dtype=vec.dtype, device=device)
-        _all_to_all_single(vec_recv, vec.reshape(-1),
-                           [B * len(m.feats_of[s]) * D for s in range(W)],
-                           [n_own * D] * W, m.group)
+        with commprobe.span("a2a_vec"):
+            _all_to_all_single(
+                vec_recv, vec.reshape(-1),
+                [B * len(m.feats_of[s]) * D for s in range(W)],
+                [n_own * D] * W, m.group)
         # assemble [B, F, D] (perm feature order) into the MLP input slice
         pos = 0
         target = deep_out_buf[:, col_offset:col_offset + m.F * D] \
@@ -186,8 +190,9 @@ class _ShardedLookup(torch.autograd.Function):
 
         wide_recv = torch.empty(W * B, dtype=wide_partial.dtype,
                                 device=device)
-        _all_to_all_single(wide_recv, wide_partial.reshape(-1),
-                           [B] * W, [B] * W, m.group)
+        with commprobe.span("a2a_wide"):
+            _all_to_all_single(wide_recv, wide_partial.reshape(-1),
+                               [B] * W, [B] * W, m.group)
         wide_out = wide_recv.reshape(W, B).sum(dim=0)
 
         ctx.module = m
@@ -224,14 +229,16 @@ class _ShardedLookup(torch.autograd.Function):
         g_send = torch.cat(sends)
         g_recv = torch.empty(W * n_own * D, dtype=g_send.dtype,
                              device=g3.device)
-        _all_to_all_single(
-            g_recv, g_send, [n_own * D] * W,
-            [B * len(m.feats_of[s]) * D for s in range(W)], m.group)
+        with commprobe.span("a2a_grad"):
+            _all_to_all_single(
+                g_recv, g_send, [n_own * D] * W,
+                [B * len(m.feats_of[s]) * D for s in range(W)], m.group)
         m._deep_sink.append((flat_ids, g_recv.reshape(W * n_own, D)))
         # wide: every owner needs every rank's grad_wide [B]
         gw = grad_wide.contiguous().reshape(-1)
         gw_all = torch.empty(W * B, dtype=gw.dtype, device=gw.device)
-        dist.all_gather_into_tensor(gw_all, gw, group=m.group)
+        with commprobe.span("ag_wide"):
+            dist.all_gather_into_tensor(gw_all, gw, group=m.group)
         m._wide_sink.append((flat_ids, gw_all))
         return None, None, None, None, None, None
 

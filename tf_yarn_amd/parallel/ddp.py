@@ -31,12 +31,15 @@ with a small first bucket, both overridable via
 from __future__ import annotations
 
 import logging
+import time
 from contextlib import contextmanager
 from typing import Dict, List, Optional, Tuple
 
 import torch
 import torch.distributed as dist
 from torch import nn
+
+from tf_yarn_amd.utils import commprobe
 
 logger = logging.getLogger(__name__)
 
@@ -260,6 +263,7 @@ class BucketedDataParallel(nn.Module):
             self._next_launch += 1
         scale_needed = (self.average and
                         dist.get_backend(self.process_group) != "nccl")
+        t0 = time.perf_counter()
         for bucket in self._buckets:
             if bucket.work is not None:
                 bucket.work.wait()
@@ -268,6 +272,8 @@ class BucketedDataParallel(nn.Module):
                 bucket.buffer.div_(self._world_size)
             bucket.pending = len(bucket.params)
             bucket.seen = [False] * len(bucket.params)
+        commprobe.add_host_time("dense_allreduce_drain",
+                                time.perf_counter() - t0)
         self._ready = [False] * len(self._buckets)
         self._next_launch = 0
 
