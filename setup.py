@@ -32,6 +32,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "bindings.cpp"),
         os.path.join(CSRC, "fused_optimizers.hip"),
         os.path.join(CSRC, "embedding.hip"),
+        os.path.join(CSRC, "binned_scatter.hip"),
         os.path.join(CSRC, "elementwise.hip"),
         os.path.join(CSRC, "wgrad.hip"),
         os.path.join(CSRC, "wgrad128.hip"),

@@ -52,6 +52,12 @@ def _world_rank(group) -> Tuple[int, int]:
 _ALLTOALL_MODE: "weakref.WeakKeyDictionary" = weakref.WeakKeyDictionary()
 
 
+def _binned_enabled() -> bool:
+    import os
+    v = os.environ.get("MIYARN_BINNED_SCATTER", "1")
+    return v != "" and v != "0"
+
+
 def _alltoall_self_check(group=None) -> str:
     """One-time probe: every rank sends rank*100+dest to dest; verify the
     native all_to_all_single delivers it.  On any mismatch or error the
@@ -346,21 +352,46 @@ class ShardedCriteoEmbeddings(nn.Module):
 
     def _apply_pending(self, lr: float) -> None:
         scale = 1.0 / self.world
+        # Binned scatter (round-2 kernel): one region-binned permutation
+        # shared by the deep and wide updates (identical flat ids), LDS
+        # dedup + exclusive-owner writeback instead of 27M random global
+        # atomics.  MIYARN_BINNED_SCATTER=0 restores the atomic kernels
+        # for A/B runs.
+        use_binned = (self.weight.is_cuda and self.dim == 16
+                      and ops.HAVE_EXT and _binned_enabled())
+        perms = {}
+
+        def perm_for(flat_ids):
+            key = flat_ids.data_ptr()
+            if key not in perms:
+                rb = ops.pick_region_bits(self.weight.shape[0],
+                                          flat_ids.numel())
+                perms[key] = ops.binned_permutation(
+                    flat_ids, self.weight.shape[0], rb)
+            return perms[key]
+
         for flat_ids, grad in self._deep_sink:
-            ops.emb_bwd_sgd(self.weight.data, flat_ids,
-                            grad.reshape(flat_ids.numel(), self.dim),
-                            lr=lr, scale=scale)
+            if use_binned:
+                ops.emb_bwd_sgd_binned(
+                    self.weight.data, flat_ids,
+                    grad.reshape(flat_ids.numel(), self.dim),
+                    lr=lr, scale=scale, perm=perm_for(flat_ids))
+            else:
+                ops.emb_bwd_sgd(self.weight.data, flat_ids,
+                                grad.reshape(flat_ids.numel(), self.dim),
+                                lr=lr, scale=scale)
         self._deep_sink.clear()
         for flat_ids, gw in self._wide_sink:
             # gw layout [W, B] (or [B] local); ids layout [peer][b][j]
-            if self.world == 1:
-                ids2d = flat_ids.reshape(gw.numel(), self.F)
-                ops.emb_scatter_sum(self.wide_weight.data, ids2d, gw,
-                                    alpha=-lr * scale)
-            else:
-                ids2d = flat_ids.reshape(gw.numel(), self.f_own)
-                ops.emb_scatter_sum(self.wide_weight.data, ids2d, gw,
-                                    alpha=-lr * scale)
+            if use_binned:
+                ops.emb_scatter_sum_binned(
+                    self.wide_weight.data, flat_ids, gw,
+                    alpha=-lr * scale, perm=perm_for(flat_ids))
+                continue
+            fan = self.F if self.world == 1 else self.f_own
+            ids2d = flat_ids.reshape(gw.numel(), fan)
+            ops.emb_scatter_sum(self.wide_weight.data, ids2d, gw,
+                                alpha=-lr * scale)
         self._wide_sink.clear()
 
     def clear_pending(self) -> None:

@@ -180,6 +180,78 @@ def emb_bwd_sgd(table: torch.Tensor, ids: torch.Tensor, grad: torch.Tensor,
                      alpha=-lr * scale)
 
 
+def pick_region_bits(n_rows: int, n_updates: int) -> int:
+    """Region size heuristic for the binned scatter: aim for an expected
+    per-bin update count around half the LDS hash capacity (512 for deep
+    tables) so dedup almost never overflows into the atomic fallback,
+    while keeping enough bins (>2048) to fill all 256 CUs."""
+    import math
+    if n_updates <= 0:
+        return 11
+    target = max(1.0, 256.0 * n_rows / n_updates)
+    bits = int(math.log2(target))
+    return max(7, min(14, bits))
+
+
+def binned_permutation(ids: torch.Tensor, n_rows: int,
+                       region_bits: int):
+    """Pass A of the binned scatter (GPU): permutation of update indices
+    grouped by 2^region_bits-row table regions + bin start offsets.
+    Algorithm spec: tests/test_binned_scatter_spec.py."""
+    _require_ext()
+    return _C.binned_permutation(ids.reshape(-1).contiguous(), n_rows,
+                                 region_bits)
+
+
+def emb_bwd_sgd_binned(table: torch.Tensor, ids: torch.Tensor,
+                       grad: torch.Tensor, *, lr: float,
+                       scale: float = 1.0,
+                       perm=None, region_bits: int = None) -> None:
+    """Binned replacement for :func:`emb_bwd_sgd` (dim-16 tables): LDS
+    hash dedup per region + exclusive-owner writeback, no global atomics
+    on the hot path.  ``perm`` = (order, starts) from
+    :func:`binned_permutation` can be shared between the deep and wide
+    tables (same flat ids).  CPU reference: plain index_add_."""
+    if not _on_gpu(table, ids, grad):
+        table.index_add_(0, ids.reshape(-1),
+                         grad.reshape(ids.numel(), -1).float(),
+                         alpha=-lr * scale)
+        return
+    _require_ext()
+    ids = ids.reshape(-1).contiguous()
+    if perm is None:
+        if region_bits is None:
+            region_bits = pick_region_bits(table.shape[0], ids.numel())
+        perm = _C.binned_permutation(ids, table.shape[0], region_bits)
+    order, starts = perm
+    _C.emb_bwd_sgd_binned(table, ids, grad.contiguous(), lr, scale,
+                          order, starts)
+
+
+def emb_scatter_sum_binned(table: torch.Tensor, ids: torch.Tensor,
+                           grad: torch.Tensor, alpha: float,
+                           perm=None, region_bits: int = None) -> None:
+    """Binned replacement for :func:`emb_scatter_sum` (scalar wide
+    tables): grad for flat update j is ``grad[j // F]``."""
+    n = ids.numel()
+    g_div = n // grad.numel()
+    if not _on_gpu(table, ids, grad):
+        expanded = grad.float().reshape(-1, 1).expand(-1, g_div) \
+            .reshape(-1)
+        table.reshape(-1).index_add_(0, ids.reshape(-1), expanded,
+                                     alpha=alpha)
+        return
+    _require_ext()
+    ids = ids.reshape(-1).contiguous()
+    if perm is None:
+        if region_bits is None:
+            region_bits = pick_region_bits(table.shape[0], n)
+        perm = _C.binned_permutation(ids, table.shape[0], region_bits)
+    order, starts = perm
+    _C.emb_scatter_sum_binned(table, ids, grad.contiguous(), g_div,
+                              alpha, order, starts)
+
+
 def emb_bwd_dense(grad_table: torch.Tensor, ids: torch.Tensor,
                   grad: torch.Tensor, scale: float = 1.0) -> None:
     if _on_gpu(grad_table, ids, grad):

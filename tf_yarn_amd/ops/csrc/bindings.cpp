@@ -46,6 +46,18 @@ void emb_fwd_into(torch::Tensor table, torch::Tensor ids,
 void emb_scatter_sum(torch::Tensor table, torch::Tensor ids,
                      torch::Tensor grad, double alpha);
 
+// binned_scatter.hip
+std::vector<torch::Tensor> binned_permutation(torch::Tensor ids,
+                                              int64_t n_rows,
+                                              int64_t region_bits);
+void emb_bwd_sgd_binned(torch::Tensor table, torch::Tensor ids,
+                        torch::Tensor grad, double lr, double scale,
+                        torch::Tensor order, torch::Tensor starts);
+void emb_scatter_sum_binned(torch::Tensor table, torch::Tensor ids,
+                            torch::Tensor grad, int64_t g_div,
+                            double alpha, torch::Tensor order,
+                            torch::Tensor starts);
+
 // elementwise.hip
 torch::Tensor bias_relu_fwd(torch::Tensor x, torch::Tensor bias);
 torch::Tensor bias_relu_bwd(torch::Tensor dy, torch::Tensor y);
@@ -108,6 +120,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Embedding gather into a slice of a larger 2D buffer");
   m.def("emb_gather_sum", &emb_gather_sum,
         "Wide-part gather-sum: out[b] = sum_f table[ids[b,f]]");
+  m.def("binned_permutation", &binned_permutation,
+        "Region-binned permutation of sparse update indices");
+  m.def("emb_bwd_sgd_binned", &emb_bwd_sgd_binned,
+        "Binned LDS-dedup scatter+SGD (deep tables, dim=16)");
+  m.def("emb_scatter_sum_binned", &emb_scatter_sum_binned,
+        "Binned LDS-dedup scatter-add (scalar wide tables)");
   m.def("emb_scatter_sum", &emb_scatter_sum,
         "Wide-part scatter: table[ids[b,f]] += alpha * g[b]");
   m.def("convert_scaled", &convert_scaled, "Scaled bf16<->fp32 convert");
