@@ -49,9 +49,15 @@ void wgrad256_kernel(const unsigned short* __restrict__ dy,
   const int64_t k_begin = (int64_t)blockIdx.y * chunk;
   const int64_t k_end = min(B, k_begin + chunk);
 
-  __shared__ __attribute__((aligned(16))) unsigned char lds_raw[2 * 256 * 128];
-  unsigned char* dyT = lds_raw;             // [256 n][64 k] swizzled
-  unsigned char* xT = lds_raw + 256 * 128;  // [256 m][64 k]
+  // Double-buffered LDS (v5): 2 x (dy 32 KB + x 32 KB) = 128 KB of the
+  // 160 KB budget.  Occupancy is VGPR-bound at 1 block/CU either way
+  // (128 fp32 acc/lane), so the second buffer is free — and it cuts the
+  // k-loop from two barriers per step to ONE: MFMAs consume buffer A
+  // while the next tile's staged registers store into buffer B.
+  __shared__ __attribute__((aligned(16)))
+      unsigned char lds_raw[2 * 2 * 256 * 128];
+  constexpr int kBufBytes = 2 * 256 * 128;
+  // dyT(b) = [256 n][64 k] swizzled; xT(b) = [256 m][64 k]
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -94,7 +100,9 @@ void wgrad256_kernel(const unsigned short* __restrict__ dy,
     }
   };
 
-  auto write_step = [&] {
+  auto write_step = [&](int buf) {
+    unsigned char* dyT = lds_raw + buf * kBufBytes;
+    unsigned char* xT = dyT + 256 * 128;
 #pragma unroll
     for (int h = 0; h < 2; ++h)
 #pragma unroll
@@ -109,12 +117,20 @@ void wgrad256_kernel(const unsigned short* __restrict__ dy,
       }
   };
 
+  // Pipeline: iteration s MFMAs buffer s&1 while the freshly loaded
+  // registers for step s+1 store into buffer (s+1)&1 — both buffers
+  // touched each step belong to different halves, so ONE barrier per
+  // step publishes the writes for the next iteration.
   load_step(k_begin);
+  write_step(0);
+  __syncthreads();
+  int buf = 0;
   for (int64_t k0 = k_begin; k0 < k_end; k0 += W4_BK) {
-    write_step();
-    __syncthreads();
-    if (k0 + W4_BK < k_end)
+    const bool has_next = (k0 + W4_BK) < k_end;
+    if (has_next)
       load_step(k0 + W4_BK);  // in flight under the MFMA phase
+    const unsigned char* dyT = lds_raw + buf * kBufBytes;
+    const unsigned char* xT = dyT + 256 * 128;
 #pragma unroll
     for (int kh = 0; kh < 2; ++kh) {
       bf16x8_v4 a[4], b[8];
@@ -133,6 +149,9 @@ void wgrad256_kernel(const unsigned short* __restrict__ dy,
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a[i], b[j], acc[i][j], 0, 0, 0);
     }
+    if (has_next)
+      write_step(buf ^ 1);
+    buf ^= 1;
     __syncthreads();
   }
 
