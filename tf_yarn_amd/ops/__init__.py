@@ -182,13 +182,13 @@ def emb_bwd_sgd(table: torch.Tensor, ids: torch.Tensor, grad: torch.Tensor,
 
 def pick_region_bits(n_rows: int, n_updates: int) -> int:
     """Region size heuristic for the binned scatter: aim for an expected
-    per-bin update count around half the LDS hash capacity (512 for deep
+    per-bin update count around half the LDS hash capacity (1024 for deep
     tables) so dedup almost never overflows into the atomic fallback,
     while keeping enough bins (>2048) to fill all 256 CUs."""
     import math
     if n_updates <= 0:
         return 11
-    target = max(1.0, 256.0 * n_rows / n_updates)
+    target = max(1.0, 512.0 * n_rows / n_updates)
     bits = int(math.log2(target))
     return max(7, min(14, bits))
 

@@ -7,25 +7,29 @@
 // at ~805 GB/s effective (340 us emb_bwd_sgd + 90 us emb_scatter_sum at
 // b=65536).  This file replaces the random atomics with:
 //
-//  Pass A (binned_permutation, 3 tiny kernels):
-//    A1 histogram ids by table region (row >> region_bits),
-//    A2 single-block exclusive scan -> bin_starts (+ cursor copy),
-//    A3 slot reservation (atomicAdd on per-bin cursors) -> `order`,
-//       a permutation of update indices grouped by region.
+//  Pass A (binned_permutation):
+//    A1 histogram ids by table region (row >> region_bits) into G=16
+//       PRIVATIZED count arrays (group = blockIdx % G) — the v1 single
+//       array serialized ~134 same-address atomics per counter and cost
+//       more than it saved (measured 178-310 us),
+//    A2 one fused kernel: per-bin group sums -> exclusive scan ->
+//       bin_starts, plus per-(group, bin) reservation cursors,
+//    A3 slot reservation through the group cursors -> `order`, a
+//       permutation of update indices grouped by region.
 //    No sort: slot order within a bin is irrelevant (sum is commutative).
 //
 //  Pass B (one workgroup per bin, grid-stride over bins):
 //    Each bin's region belongs to EXACTLY ONE workgroup, so the final
 //    table update needs no global atomics at all:
 //      1. dedup/accumulate the bin's updates in an LDS hash keyed by
-//         row id (LDS atomicCAS insert + LDS atomicAdd accumulate);
+//         row id (LDS atomicCAS insert + LDS float atomicAdd, value
+//         stride PADDED to 17 — the natural *16 stride made every
+//         d-loop access a 16-way bank conflict);
 //      2. barrier; write each occupied hash slot back with a plain
-//         vectorized read-modify-write.
-//    Traffic drops from `updates x 128 B` of random atomics to
-//    `unique_rows x 128 B` of sequential-ish RMW + the grad reads.
-//    Bins whose update count exceeds the hash capacity (tiny tables,
-//    heavy skew) fall back to direct global atomics — still
-//    region-grouped, so they keep the L2/DRAM-row locality win.
+//         read-modify-write.
+//    Bins whose update count exceeds 3/4 of the hash capacity (tiny
+//    tables, heavy skew) fall back to direct global atomics — still
+//    region-grouped, so they keep the L2 locality.
 //
 // The deep (dim=16) and wide (dim=1) tables of the CTR models use the
 // SAME flat ids, so the host computes one permutation and feeds both
@@ -37,63 +41,73 @@
 
 namespace {
 
+constexpr int kGroups = 16;       // pass-A privatization factor
 constexpr int kScanBlock = 1024;
-constexpr int kHashDeep = 512;    // slots; 512*(4+64) B = 34 KB LDS
+constexpr int kHashDeep = 1024;   // slots; 1024*(4 + 17*4) B = 72 KB LDS
+constexpr int kValStride = 17;    // bank-conflict-free value stride
 constexpr int kHashScalar = 2048; // slots; 2048*(4+4) B = 16 KB LDS
 constexpr int kProbeMax = 64;
 
 __global__ void bin_count_kernel(const int64_t* __restrict__ ids,
                                  int64_t n, int32_t* __restrict__ counts,
-                                 int region_bits) {
+                                 int n_bins, int region_bits) {
+  int32_t* my = counts + (blockIdx.x % kGroups) * (int64_t)n_bins;
   const int64_t stride = gridDim.x * (int64_t)blockDim.x;
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
        i < n; i += stride)
-    atomicAdd(&counts[ids[i] >> region_bits], 1);
+    atomicAdd(&my[ids[i] >> region_bits], 1);
 }
 
-// Single-block exclusive scan: counts[n_bins] -> starts[n_bins+1]; also
-// primes the per-bin reservation cursors.  n_bins is at most a few 100K
-// (rows >> 7), so one 1024-thread block in chunks is microseconds.
+// Fused single-block pass: per-bin totals over the G group counts,
+// exclusive scan -> starts[n_bins+1], and per-(group,bin) cursors
+// (cursor[g][bin] = starts[bin] + sum of groups < g).  Threads own
+// CONTIGUOUS bin chunks so the scan is chunk-total scan + local walk.
 __global__ void bin_scan_kernel(const int32_t* __restrict__ counts,
                                 int32_t* __restrict__ starts,
                                 int32_t* __restrict__ cursor, int n_bins) {
-  __shared__ int32_t tmp[kScanBlock];
-  __shared__ int32_t running;
-  if (threadIdx.x == 0) {
-    running = 0;
-    starts[0] = 0;
+  __shared__ int32_t chunk_tot[kScanBlock];
+  const int per = (n_bins + kScanBlock - 1) / kScanBlock;
+  const int b0 = threadIdx.x * per;
+  const int b1 = min(n_bins, b0 + per);
+  int32_t local = 0;
+  for (int b = b0; b < b1; ++b) {
+    int32_t t = 0;
+#pragma unroll 4
+    for (int g = 0; g < kGroups; ++g) t += counts[g * (int64_t)n_bins + b];
+    local += t;
   }
+  chunk_tot[threadIdx.x] = local;
   __syncthreads();
-  for (int base = 0; base < n_bins; base += kScanBlock) {
-    const int i = base + threadIdx.x;
-    const int32_t v = (i < n_bins) ? counts[i] : 0;
-    tmp[threadIdx.x] = v;
+  // exclusive scan of the 1024 chunk totals
+  for (int off = 1; off < kScanBlock; off <<= 1) {
+    int32_t t = (threadIdx.x >= off) ? chunk_tot[threadIdx.x - off] : 0;
     __syncthreads();
-    for (int off = 1; off < kScanBlock; off <<= 1) {
-      int32_t t = (threadIdx.x >= off) ? tmp[threadIdx.x - off] : 0;
-      __syncthreads();
-      tmp[threadIdx.x] += t;
-      __syncthreads();
+    chunk_tot[threadIdx.x] += t;
+    __syncthreads();
+  }
+  int32_t run = (threadIdx.x == 0) ? 0 : chunk_tot[threadIdx.x - 1];
+  if (threadIdx.x == 0) starts[0] = 0;
+  for (int b = b0; b < b1; ++b) {
+    int32_t acc = run;
+#pragma unroll 4
+    for (int g = 0; g < kGroups; ++g) {
+      cursor[g * (int64_t)n_bins + b] = acc;
+      acc += counts[g * (int64_t)n_bins + b];
     }
-    if (i < n_bins) {
-      const int32_t excl = running + tmp[threadIdx.x] - v;
-      cursor[i] = excl;
-      starts[i + 1] = running + tmp[threadIdx.x];
-    }
-    __syncthreads();
-    if (threadIdx.x == 0) running += tmp[kScanBlock - 1];
-    __syncthreads();
+    run = acc;
+    starts[b + 1] = run;
   }
 }
 
 __global__ void bin_slot_kernel(const int64_t* __restrict__ ids,
                                 int64_t n, int32_t* __restrict__ cursor,
                                 int32_t* __restrict__ order,
-                                int region_bits) {
+                                int n_bins, int region_bits) {
+  int32_t* my = cursor + (blockIdx.x % kGroups) * (int64_t)n_bins;
   const int64_t stride = gridDim.x * (int64_t)blockDim.x;
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
        i < n; i += stride) {
-    const int slot = atomicAdd(&cursor[ids[i] >> region_bits], 1);
+    const int slot = atomicAdd(&my[ids[i] >> region_bits], 1);
     order[slot] = static_cast<int32_t>(i);
   }
 }
@@ -127,7 +141,7 @@ __global__ void binned_apply_deep_kernel(
     const int32_t* __restrict__ order,
     const int32_t* __restrict__ starts, int n_bins, float neg_lr_scale) {
   __shared__ int32_t h_id[kHashDeep];
-  __shared__ float h_val[kHashDeep * DIM];
+  __shared__ float h_val[kHashDeep * kValStride];
   constexpr int DVEC = DIM / 4;
   for (int bin = blockIdx.x; bin < n_bins; bin += gridDim.x) {
     const int start = starts[bin], end = starts[bin + 1];
@@ -135,10 +149,11 @@ __global__ void binned_apply_deep_kernel(
     if (count == 0) continue;
     const bool use_hash = count <= (3 * kHashDeep) / 4;
     if (use_hash) {
-      for (int s = threadIdx.x; s < kHashDeep; s += blockDim.x)
+      for (int s = threadIdx.x; s < kHashDeep; s += blockDim.x) {
         h_id[s] = -1;
-      for (int s = threadIdx.x; s < kHashDeep * DIM; s += blockDim.x)
-        h_val[s] = 0.f;
+#pragma unroll
+        for (int d = 0; d < DIM; ++d) h_val[s * kValStride + d] = 0.f;
+      }
       __syncthreads();
     }
     for (int k = start + threadIdx.x; k < end; k += blockDim.x) {
@@ -159,7 +174,7 @@ __global__ void binned_apply_deep_kernel(
       if (slot >= 0) {
 #pragma unroll
         for (int d = 0; d < DIM; ++d)
-          atomicAdd(&h_val[slot * DIM + d], gv[d]);
+          atomicAdd(&h_val[slot * kValStride + d], gv[d]);
       } else {
         float* dst = table + row * DIM;
 #pragma unroll
@@ -178,7 +193,7 @@ __global__ void binned_apply_deep_kernel(
           f32x4 cur = reinterpret_cast<f32x4*>(dst)[q];
 #pragma unroll
           for (int x = 0; x < 4; ++x)
-            cur[x] += neg_lr_scale * h_val[s * DIM + q * 4 + x];
+            cur[x] += neg_lr_scale * h_val[s * kValStride + q * 4 + x];
           reinterpret_cast<f32x4*>(dst)[q] = cur;
         }
       }
@@ -250,15 +265,15 @@ std::vector<torch::Tensor> binned_permutation(torch::Tensor ids,
   TORCH_CHECK(n_bins64 < (1 << 24), "too many bins; raise region_bits");
   const int n_bins = static_cast<int>(n_bins64);
   auto opts = ids.options().dtype(torch::kInt32);
-  auto counts = torch::zeros({n_bins}, opts);
+  auto counts = torch::zeros({kGroups, n_bins}, opts);
   auto starts = torch::empty({n_bins + 1}, opts);
-  auto cursor = torch::empty({n_bins}, opts);
+  auto cursor = torch::empty({kGroups, n_bins}, opts);
   auto order = torch::empty({n}, opts);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int grid = miyarn_grid(n);
   hipLaunchKernelGGL(bin_count_kernel, dim3(grid), dim3(MIYARN_BLOCK), 0,
                      stream, ids.data_ptr<int64_t>(), n,
-                     counts.data_ptr<int32_t>(),
+                     counts.data_ptr<int32_t>(), n_bins,
                      static_cast<int>(region_bits));
   hipLaunchKernelGGL(bin_scan_kernel, dim3(1), dim3(kScanBlock), 0,
                      stream, counts.data_ptr<int32_t>(),
@@ -267,7 +282,7 @@ std::vector<torch::Tensor> binned_permutation(torch::Tensor ids,
   hipLaunchKernelGGL(bin_slot_kernel, dim3(grid), dim3(MIYARN_BLOCK), 0,
                      stream, ids.data_ptr<int64_t>(), n,
                      cursor.data_ptr<int32_t>(),
-                     order.data_ptr<int32_t>(),
+                     order.data_ptr<int32_t>(), n_bins,
                      static_cast<int>(region_bits));
   return {order, starts};
 }
