@@ -11,7 +11,8 @@ import time
 
 import torch
 
-sys.path.insert(0, ".")
+import os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from tf_yarn_amd import ops  # noqa: E402
 
 B = 65536

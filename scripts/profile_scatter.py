@@ -7,7 +7,8 @@ import sys
 
 import torch
 
-sys.path.insert(0, ".")
+import os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from tf_yarn_amd import ops  # noqa: E402
 
 B, F, ROWS_PER, DIM = 65536, 26, 1_000_000, 16
