@@ -681,11 +681,15 @@ def test_binned_permutation_matches_spec():
     n_rows, n_upd, bits = 300_000, 50_000, 12
     ids = torch.randint(0, n_rows, (n_upd,), device="cuda")
     order, starts = ops.binned_permutation(ids, n_rows, bits)
-    order, starts = order.cpu().long(), starts.cpu().long()
+    # order packs (row << 31 | update index)
+    j = (order & ((1 << 31) - 1)).cpu()
+    rows = (order >> 31).cpu()
+    starts = starts.cpu().long()
     ids_cpu = ids.cpu()
-    assert sorted(order.tolist()) == list(range(n_upd))
+    assert sorted(j.tolist()) == list(range(n_upd))
+    assert (ids_cpu[j] == rows).all()  # packed rows match the ids
     for b in range(starts.numel() - 1):
-        sl = order[starts[b]:starts[b + 1]]
+        sl = j[starts[b]:starts[b + 1]]
         if sl.numel():
             assert ((ids_cpu[sl] >> bits) == b).all()
 

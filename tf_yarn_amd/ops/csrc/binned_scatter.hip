@@ -5,28 +5,32 @@
 //
 // Round-1 floor: 27M random fp32 atomicAdds/step over a 1.7 GB table ran
 // at ~805 GB/s effective (340 us emb_bwd_sgd + 90 us emb_scatter_sum at
-// b=65536).  This file replaces the random atomics with:
+// b=65536).  Design:
 //
 //  Pass A (binned_permutation):
-//    A1 histogram ids by table region (row >> region_bits) into G=16
-//       PRIVATIZED count arrays (group = blockIdx % G) — the v1 single
-//       array serialized ~134 same-address atomics per counter and cost
-//       more than it saved (measured 178-310 us),
-//    A2 one fused kernel: per-bin group sums -> exclusive scan ->
-//       bin_starts, plus per-(group, bin) reservation cursors,
-//    A3 slot reservation through the group cursors -> `order`, a
-//       permutation of update indices grouped by region.
-//    No sort: slot order within a bin is irrelevant (sum is commutative).
+//    A1 histogram ids by table region (row >> region_bits).  Counters
+//       are padded to ONE PER 64 B CACHE LINE: v2's 16-way privatized
+//       dense counters still serialized at L2 line granularity
+//       (measured 65 us for 1.7M int adds; ~134 atomics landed on each
+//       line).
+//    A2 single-block exclusive scan -> bin_starts + reservation
+//       cursors (G=1 after padding, so the scan walks n_bins once),
+//    A3 slot reservation; order[slot] packs (row << 31 | update index)
+//       so pass B never touches the ids array again (the v2 applies
+//       paid a random 8 B ids[j] load per update).
+//    No sort: slot order within a bin is irrelevant (sum is
+//    commutative).
 //
 //  Pass B (one workgroup per bin, grid-stride over bins):
 //    Each bin's region belongs to EXACTLY ONE workgroup, so the final
 //    table update needs no global atomics at all:
 //      1. dedup/accumulate the bin's updates in an LDS hash keyed by
 //         row id (LDS atomicCAS insert + LDS float atomicAdd, value
-//         stride PADDED to 17 — the natural *16 stride made every
-//         d-loop access a 16-way bank conflict);
+//         stride padded to 17 — the natural *16 stride is a 16-way
+//         bank conflict);
 //      2. barrier; write each occupied hash slot back with a plain
-//         read-modify-write.
+//         vectorized read-modify-write, resetting the slot lazily for
+//         the block's next bin.
 //    Bins whose update count exceeds 3/4 of the hash capacity (tiny
 //    tables, heavy skew) fall back to direct global atomics — still
 //    region-grouped, so they keep the L2 locality.
@@ -41,27 +45,26 @@
 
 namespace {
 
-constexpr int kGroups = 16;       // pass-A privatization factor
+constexpr int kPad = 16;          // int32 counters, one per 64 B line
 constexpr int kScanBlock = 1024;
 constexpr int kHashDeep = 1024;   // slots; 1024*(4 + 17*4) B = 72 KB LDS
 constexpr int kValStride = 17;    // bank-conflict-free value stride
 constexpr int kHashScalar = 2048; // slots; 2048*(4+4) B = 16 KB LDS
 constexpr int kProbeMax = 64;
+constexpr int kJBits = 31;
+constexpr int64_t kJMask = (int64_t{1} << kJBits) - 1;
 
 __global__ void bin_count_kernel(const int64_t* __restrict__ ids,
                                  int64_t n, int32_t* __restrict__ counts,
-                                 int n_bins, int region_bits) {
-  int32_t* my = counts + (blockIdx.x % kGroups) * (int64_t)n_bins;
+                                 int region_bits) {
   const int64_t stride = gridDim.x * (int64_t)blockDim.x;
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
        i < n; i += stride)
-    atomicAdd(&my[ids[i] >> region_bits], 1);
+    atomicAdd(&counts[(ids[i] >> region_bits) * kPad], 1);
 }
 
-// Fused single-block pass: per-bin totals over the G group counts,
-// exclusive scan -> starts[n_bins+1], and per-(group,bin) cursors
-// (cursor[g][bin] = starts[bin] + sum of groups < g).  Threads own
-// CONTIGUOUS bin chunks so the scan is chunk-total scan + local walk.
+// Single-block exclusive scan over the line-padded counters ->
+// starts[n_bins+1] and the (padded) reservation cursors.
 __global__ void bin_scan_kernel(const int32_t* __restrict__ counts,
                                 int32_t* __restrict__ starts,
                                 int32_t* __restrict__ cursor, int n_bins) {
@@ -70,15 +73,9 @@ __global__ void bin_scan_kernel(const int32_t* __restrict__ counts,
   const int b0 = threadIdx.x * per;
   const int b1 = min(n_bins, b0 + per);
   int32_t local = 0;
-  for (int b = b0; b < b1; ++b) {
-    int32_t t = 0;
-#pragma unroll 4
-    for (int g = 0; g < kGroups; ++g) t += counts[g * (int64_t)n_bins + b];
-    local += t;
-  }
+  for (int b = b0; b < b1; ++b) local += counts[(int64_t)b * kPad];
   chunk_tot[threadIdx.x] = local;
   __syncthreads();
-  // exclusive scan of the 1024 chunk totals
   for (int off = 1; off < kScanBlock; off <<= 1) {
     int32_t t = (threadIdx.x >= off) ? chunk_tot[threadIdx.x - off] : 0;
     __syncthreads();
@@ -88,27 +85,22 @@ __global__ void bin_scan_kernel(const int32_t* __restrict__ counts,
   int32_t run = (threadIdx.x == 0) ? 0 : chunk_tot[threadIdx.x - 1];
   if (threadIdx.x == 0) starts[0] = 0;
   for (int b = b0; b < b1; ++b) {
-    int32_t acc = run;
-#pragma unroll 4
-    for (int g = 0; g < kGroups; ++g) {
-      cursor[g * (int64_t)n_bins + b] = acc;
-      acc += counts[g * (int64_t)n_bins + b];
-    }
-    run = acc;
+    cursor[(int64_t)b * kPad] = run;
+    run += counts[(int64_t)b * kPad];
     starts[b + 1] = run;
   }
 }
 
 __global__ void bin_slot_kernel(const int64_t* __restrict__ ids,
                                 int64_t n, int32_t* __restrict__ cursor,
-                                int32_t* __restrict__ order,
-                                int n_bins, int region_bits) {
-  int32_t* my = cursor + (blockIdx.x % kGroups) * (int64_t)n_bins;
+                                int64_t* __restrict__ order,
+                                int region_bits) {
   const int64_t stride = gridDim.x * (int64_t)blockDim.x;
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
        i < n; i += stride) {
-    const int slot = atomicAdd(&my[ids[i] >> region_bits], 1);
-    order[slot] = static_cast<int32_t>(i);
+    const int64_t row = ids[i];
+    const int slot = atomicAdd(&cursor[(row >> region_bits) * kPad], 1);
+    order[slot] = (row << kJBits) | i;
   }
 }
 
@@ -136,16 +128,15 @@ __device__ __forceinline__ int hash_find_or_insert(int32_t* h_id,
 // Pass B, deep tables (dim compile-time, 16 for the CTR models).
 template <typename GIo, int DIM>
 __global__ void binned_apply_deep_kernel(
-    float* __restrict__ table, const int64_t* __restrict__ ids,
+    float* __restrict__ table,
     const typename GIo::scalar_t* __restrict__ g,
-    const int32_t* __restrict__ order,
+    const int64_t* __restrict__ order,
     const int32_t* __restrict__ starts, int n_bins, float neg_lr_scale) {
   __shared__ int32_t h_id[kHashDeep];
   __shared__ float h_val[kHashDeep * kValStride];
   constexpr int DVEC = DIM / 4;
   // Init the hash ONCE; each bin's writeback resets only the slots it
-  // occupied (v2 re-zeroed all 1024x17 words per bin — 110M LDS writes
-  // across 6.3K bins, 4x the useful atomic work).
+  // occupied.
   for (int s = threadIdx.x; s < kHashDeep; s += blockDim.x) {
     h_id[s] = -1;
 #pragma unroll
@@ -158,13 +149,14 @@ __global__ void binned_apply_deep_kernel(
     if (count == 0) continue;
     const bool use_hash = count <= (3 * kHashDeep) / 4;
     for (int k = start + threadIdx.x; k < end; k += blockDim.x) {
-      const int j = order[k];
-      const int64_t row = ids[j];
+      const int64_t packed = order[k];
+      const int64_t j = packed & kJMask;
+      const int64_t row = packed >> kJBits;
       float gv[DIM];
 #pragma unroll
       for (int q = 0; q < DVEC; ++q) {
         float quad[4];
-        QuadIo<GIo>::load4(g, (int64_t)j * DVEC + q, quad);
+        QuadIo<GIo>::load4(g, j * DVEC + q, quad);
 #pragma unroll
         for (int x = 0; x < 4; ++x) gv[q * 4 + x] = quad[x];
       }
@@ -210,9 +202,9 @@ __global__ void binned_apply_deep_kernel(
 // (g_div = features-per-output of the fused gather-sum).
 template <typename GIo>
 __global__ void binned_apply_scalar_kernel(
-    float* __restrict__ table, const int64_t* __restrict__ ids,
+    float* __restrict__ table,
     const typename GIo::scalar_t* __restrict__ g,
-    const int32_t* __restrict__ order,
+    const int64_t* __restrict__ order,
     const int32_t* __restrict__ starts, int n_bins, int g_div,
     float alpha) {
   __shared__ int32_t h_id[kHashScalar];
@@ -228,8 +220,9 @@ __global__ void binned_apply_scalar_kernel(
     if (count == 0) continue;
     const bool use_hash = count <= (3 * kHashScalar) / 4;
     for (int k = start + threadIdx.x; k < end; k += blockDim.x) {
-      const int j = order[k];
-      const int64_t row = ids[j];
+      const int64_t packed = order[k];
+      const int64_t j = packed & kJMask;
+      const int64_t row = packed >> kJBits;
       const float gv = GIo::load(g, j / g_div);
       int slot = -1;
       if (use_hash)
@@ -266,20 +259,21 @@ std::vector<torch::Tensor> binned_permutation(torch::Tensor ids,
   TORCH_CHECK(region_bits >= 1 && region_bits <= 26, "bad region_bits");
   TORCH_CHECK(n_rows < (int64_t{1} << 31), "table too large for int32 rows");
   const int64_t n = ids.numel();
+  TORCH_CHECK(n < (int64_t{1} << kJBits), "too many updates to pack");
   const int64_t n_bins64 =
       (n_rows + (int64_t{1} << region_bits) - 1) >> region_bits;
   TORCH_CHECK(n_bins64 < (1 << 24), "too many bins; raise region_bits");
   const int n_bins = static_cast<int>(n_bins64);
   auto opts = ids.options().dtype(torch::kInt32);
-  auto counts = torch::zeros({kGroups, n_bins}, opts);
+  auto counts = torch::zeros({n_bins * kPad}, opts);
   auto starts = torch::empty({n_bins + 1}, opts);
-  auto cursor = torch::empty({kGroups, n_bins}, opts);
-  auto order = torch::empty({n}, opts);
+  auto cursor = torch::empty({n_bins * kPad}, opts);
+  auto order = torch::empty({n}, ids.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int grid = miyarn_grid(n);
   hipLaunchKernelGGL(bin_count_kernel, dim3(grid), dim3(MIYARN_BLOCK), 0,
                      stream, ids.data_ptr<int64_t>(), n,
-                     counts.data_ptr<int32_t>(), n_bins,
+                     counts.data_ptr<int32_t>(),
                      static_cast<int>(region_bits));
   hipLaunchKernelGGL(bin_scan_kernel, dim3(1), dim3(kScanBlock), 0,
                      stream, counts.data_ptr<int32_t>(),
@@ -288,7 +282,7 @@ std::vector<torch::Tensor> binned_permutation(torch::Tensor ids,
   hipLaunchKernelGGL(bin_slot_kernel, dim3(grid), dim3(MIYARN_BLOCK), 0,
                      stream, ids.data_ptr<int64_t>(), n,
                      cursor.data_ptr<int32_t>(),
-                     order.data_ptr<int32_t>(), n_bins,
+                     order.data_ptr<int64_t>(),
                      static_cast<int>(region_bits));
   return {order, starts};
 }
@@ -304,7 +298,7 @@ void emb_bwd_sgd_binned(torch::Tensor table, torch::Tensor ids,
               "); use emb_bwd_sgd");
   TORCH_CHECK(grad.is_cuda() && grad.is_contiguous() &&
               grad.numel() == n * dim, "grad shape mismatch");
-  TORCH_CHECK(order.scalar_type() == torch::kInt32 &&
+  TORCH_CHECK(order.scalar_type() == torch::kInt64 &&
               starts.scalar_type() == torch::kInt32, "bad perm dtypes");
   const int n_bins = static_cast<int>(starts.numel() - 1);
   auto stream = c10::hip::getCurrentHIPStream().stream();
@@ -313,16 +307,16 @@ void emb_bwd_sgd_binned(torch::Tensor table, torch::Tensor ids,
   if (grad.scalar_type() == torch::kFloat32) {
     hipLaunchKernelGGL((binned_apply_deep_kernel<F32Io, 16>), dim3(grid),
                        dim3(MIYARN_BLOCK), 0, stream,
-                       table.data_ptr<float>(), ids.data_ptr<int64_t>(),
-                       grad.data_ptr<float>(), order.data_ptr<int32_t>(),
+                       table.data_ptr<float>(),
+                       grad.data_ptr<float>(), order.data_ptr<int64_t>(),
                        starts.data_ptr<int32_t>(), n_bins, nls);
   } else {
     TORCH_CHECK(grad.scalar_type() == torch::kBFloat16, "fp32/bf16 only");
     hipLaunchKernelGGL((binned_apply_deep_kernel<Bf16Io, 16>), dim3(grid),
                        dim3(MIYARN_BLOCK), 0, stream,
-                       table.data_ptr<float>(), ids.data_ptr<int64_t>(),
+                       table.data_ptr<float>(),
                        reinterpret_cast<unsigned short*>(grad.data_ptr()),
-                       order.data_ptr<int32_t>(),
+                       order.data_ptr<int64_t>(),
                        starts.data_ptr<int32_t>(), n_bins, nls);
   }
 }
@@ -336,7 +330,7 @@ void emb_scatter_sum_binned(torch::Tensor table, torch::Tensor ids,
               table.scalar_type() == torch::kFloat32, "bad table");
   TORCH_CHECK(grad.is_cuda() && grad.is_contiguous() &&
               grad.numel() * g_div == n, "grad/g_div mismatch");
-  TORCH_CHECK(order.scalar_type() == torch::kInt32 &&
+  TORCH_CHECK(order.scalar_type() == torch::kInt64 &&
               starts.scalar_type() == torch::kInt32, "bad perm dtypes");
   const int n_bins = static_cast<int>(starts.numel() - 1);
   auto stream = c10::hip::getCurrentHIPStream().stream();
@@ -344,8 +338,8 @@ void emb_scatter_sum_binned(torch::Tensor table, torch::Tensor ids,
   if (grad.scalar_type() == torch::kFloat32) {
     hipLaunchKernelGGL(binned_apply_scalar_kernel<F32Io>, dim3(grid),
                        dim3(MIYARN_BLOCK), 0, stream,
-                       table.data_ptr<float>(), ids.data_ptr<int64_t>(),
-                       grad.data_ptr<float>(), order.data_ptr<int32_t>(),
+                       table.data_ptr<float>(),
+                       grad.data_ptr<float>(), order.data_ptr<int64_t>(),
                        starts.data_ptr<int32_t>(), n_bins,
                        static_cast<int>(g_div),
                        static_cast<float>(alpha));
@@ -353,9 +347,9 @@ void emb_scatter_sum_binned(torch::Tensor table, torch::Tensor ids,
     TORCH_CHECK(grad.scalar_type() == torch::kBFloat16, "fp32/bf16 only");
     hipLaunchKernelGGL(binned_apply_scalar_kernel<Bf16Io>, dim3(grid),
                        dim3(MIYARN_BLOCK), 0, stream,
-                       table.data_ptr<float>(), ids.data_ptr<int64_t>(),
+                       table.data_ptr<float>(),
                        reinterpret_cast<unsigned short*>(grad.data_ptr()),
-                       order.data_ptr<int32_t>(),
+                       order.data_ptr<int64_t>(),
                        starts.data_ptr<int32_t>(), n_bins,
                        static_cast<int>(g_div),
                        static_cast<float>(alpha));
