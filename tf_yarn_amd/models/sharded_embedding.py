@@ -151,6 +151,7 @@ class _ShardedLookup(torch.autograd.Function):
             ops.emb_fwd_into(deep_table, flat, deep_out_buf, col_offset)
             wide_out = ops.emb_gather_sum(wide_table, flat.reshape(B, -1),
                                           m.out_bf16)
+            m._start_perm_async(flat)
             ctx.module = m
             ctx.save_for_backward(flat)
             ctx.local_only = True
@@ -169,6 +170,8 @@ class _ShardedLookup(torch.autograd.Function):
                                m.group)
         # local row offsets: layout [peer][b][j], j = feature slot
         flat_local = ids_recv + m.own_offsets_tiled[:ids_recv.numel()]
+
+        m._start_perm_async(flat_local)
 
         # ---- owner-local gathers ----------------------------------------
         vec = ops.emb_fwd(deep_table, flat_local, m.out_bf16)  # [W*n_own, D]
@@ -302,6 +305,8 @@ class ShardedCriteoEmbeddings(nn.Module):
         self._deep_sink: List[Tuple[torch.Tensor, torch.Tensor]] = []
         self._wide_sink: List[Tuple[torch.Tensor, torch.Tensor]] = []
         self._last_col_offset = 0
+        self._perm_stream = None  # side stream for async pass A
+        self._pending_perm = None  # (ids data_ptr, (order, starts), event)
 
     def _ensure_tiled(self, numel: int, device) -> None:
         if self.own_offsets_tiled.numel() < numel:
@@ -350,6 +355,26 @@ class ShardedCriteoEmbeddings(nn.Module):
             return
         self._apply_pending(lr)
 
+    def _start_perm_async(self, flat_ids: torch.Tensor) -> None:
+        """Kick off the binned-scatter permutation (pass A) on a side
+        stream DURING FORWARD: it depends only on the ids, and its
+        ~200 us of latency/atomic-bound work hides under the MFMA-bound
+        MLP GEMMs instead of sitting on the backward critical path."""
+        if not (flat_ids.is_cuda and self.dim == 16 and ops.HAVE_EXT
+                and torch.is_grad_enabled() and _binned_enabled()):
+            return
+        rb = ops.pick_region_bits(self.weight.shape[0], flat_ids.numel())
+        if self._perm_stream is None:
+            self._perm_stream = torch.cuda.Stream()
+        self._perm_stream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(self._perm_stream):
+            flat_ids.record_stream(self._perm_stream)
+            perm = ops.binned_permutation(flat_ids, self.weight.shape[0],
+                                          rb)
+        ev = torch.cuda.Event()
+        ev.record(self._perm_stream)
+        self._pending_perm = (flat_ids.data_ptr(), perm, ev)
+
     def _apply_pending(self, lr: float) -> None:
         scale = 1.0 / self.world
         # Binned scatter (round-2 kernel): one region-binned permutation
@@ -360,6 +385,13 @@ class ShardedCriteoEmbeddings(nn.Module):
         use_binned = (self.weight.is_cuda and self.dim == 16
                       and ops.HAVE_EXT and _binned_enabled())
         perms = {}
+        if use_binned and self._pending_perm is not None:
+            key, perm, ev = self._pending_perm
+            torch.cuda.current_stream().wait_event(ev)
+            for t in perm:
+                t.record_stream(torch.cuda.current_stream())
+            perms[key] = perm
+            self._pending_perm = None
 
         def perm_for(flat_ids):
             key = flat_ids.data_ptr()
