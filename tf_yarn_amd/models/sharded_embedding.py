@@ -58,6 +58,12 @@ def _binned_enabled() -> bool:
     return v != "" and v != "0"
 
 
+def _perm_stream_enabled() -> bool:
+    import os
+    v = os.environ.get("MIYARN_PERM_STREAM", "1")
+    return v != "" and v != "0"
+
+
 def _alltoall_self_check(group=None) -> str:
     """One-time probe: every rank sends rank*100+dest to dest; verify the
     native all_to_all_single delivers it.  On any mismatch or error the
@@ -364,6 +370,12 @@ class ShardedCriteoEmbeddings(nn.Module):
                 and torch.is_grad_enabled() and _binned_enabled()):
             return
         rb = ops.pick_region_bits(self.weight.shape[0], flat_ids.numel())
+        if not _perm_stream_enabled():
+            # inline (A/B diagnostic): pass A runs on the main stream
+            perm = ops.binned_permutation(flat_ids, self.weight.shape[0],
+                                          rb)
+            self._pending_perm = (flat_ids.data_ptr(), perm, None)
+            return
         if self._perm_stream is None:
             self._perm_stream = torch.cuda.Stream()
         self._perm_stream.wait_stream(torch.cuda.current_stream())
@@ -387,9 +399,10 @@ class ShardedCriteoEmbeddings(nn.Module):
         perms = {}
         if use_binned and self._pending_perm is not None:
             key, perm, ev = self._pending_perm
-            torch.cuda.current_stream().wait_event(ev)
-            for t in perm:
-                t.record_stream(torch.cuda.current_stream())
+            if ev is not None:
+                torch.cuda.current_stream().wait_event(ev)
+                for t in perm:
+                    t.record_stream(torch.cuda.current_stream())
             perms[key] = perm
             self._pending_perm = None
 
