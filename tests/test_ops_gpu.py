@@ -719,3 +719,24 @@ def test_sharded_embedding_binned_vs_atomic_path():
     os.environ.pop("MIYARN_BINNED_SCATTER", None)
     assert torch.allclose(results["1"][0], results["0"][0], atol=1e-4)
     assert torch.allclose(results["1"][1], results["0"][1], atol=1e-4)
+
+
+@requires_gpu
+@pytest.mark.parametrize("gdtype", [torch.float32, torch.bfloat16])
+def test_emb_bwd_sgd_fused_wide_matches_separate(gdtype):
+    torch.manual_seed(8)
+    n_rows, batch, fan, dim = 200_000, 10_000, 13, 16
+    table = torch.randn(n_rows, dim, device="cuda")
+    wide = torch.randn(n_rows, 1, device="cuda")
+    ref_t, ref_w = table.clone(), wide.clone()
+    ids = torch.randint(0, n_rows, (batch * fan,), device="cuda")
+    grad = torch.randn(batch * fan, dim, device="cuda").to(gdtype)
+    gw = torch.randn(batch, device="cuda").to(gdtype)
+    ops.emb_bwd_sgd_fused_wide(table, wide, ids, grad, gw,
+                               lr=0.1, scale=0.5)
+    ref_t.index_add_(0, ids, grad.float(), alpha=-0.05)
+    expanded = gw.float().reshape(-1, 1).expand(-1, fan).reshape(-1)
+    ref_w.reshape(-1).index_add_(0, ids, expanded, alpha=-0.05)
+    atol = 1e-4 if gdtype == torch.float32 else 2e-2
+    assert torch.allclose(table, ref_t, atol=atol)
+    assert torch.allclose(wide, ref_w, atol=atol)

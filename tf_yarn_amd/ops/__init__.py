@@ -252,6 +252,29 @@ def emb_scatter_sum_binned(table: torch.Tensor, ids: torch.Tensor,
                               alpha, order, starts)
 
 
+def emb_bwd_sgd_fused_wide(table: torch.Tensor, wide_table: torch.Tensor,
+                           ids: torch.Tensor, grad: torch.Tensor,
+                           gw: torch.Tensor, *, lr: float,
+                           scale: float = 1.0) -> None:
+    """Fused sparse update of BOTH CTR tables from one pass over the
+    shared flat ids:  ``table[ids[i]] -= lr*scale*grad[i]`` and
+    ``wide_table[ids[i]] -= lr*scale*gw[i // (n//gw.numel())]``."""
+    if _on_gpu(table, ids, grad) and grad.dtype == gw.dtype:
+        _require_ext()
+        _C.emb_bwd_sgd_fused_wide(table, wide_table,
+                                  ids.reshape(-1).contiguous(),
+                                  grad.contiguous(), gw.contiguous(),
+                                  lr, scale)
+        return
+    n = ids.numel()
+    g_div = n // gw.numel()
+    table.index_add_(0, ids.reshape(-1),
+                     grad.reshape(n, -1).float(), alpha=-lr * scale)
+    expanded = gw.float().reshape(-1, 1).expand(-1, g_div).reshape(-1)
+    wide_table.reshape(-1).index_add_(0, ids.reshape(-1), expanded,
+                                      alpha=-lr * scale)
+
+
 def emb_bwd_dense(grad_table: torch.Tensor, ids: torch.Tensor,
                   grad: torch.Tensor, scale: float = 1.0) -> None:
     if _on_gpu(grad_table, ids, grad):

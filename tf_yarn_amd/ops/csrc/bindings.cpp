@@ -38,6 +38,9 @@ void emb_bwd_sgd_sorted(torch::Tensor table, torch::Tensor sorted_ids,
                         torch::Tensor grad, double lr, double scale);
 void emb_bwd_dense(torch::Tensor grad_table, torch::Tensor ids,
                    torch::Tensor grad, double scale);
+void emb_bwd_sgd_fused_wide(torch::Tensor table, torch::Tensor wide_table,
+                            torch::Tensor ids, torch::Tensor grad,
+                            torch::Tensor gw, double lr, double scale);
 
 torch::Tensor emb_gather_sum(torch::Tensor table, torch::Tensor ids,
                              int64_t batch, bool out_bf16);
@@ -96,6 +99,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused sparse embedding grad scatter + SGD update");
   m.def("emb_bwd_sgd_sorted", &emb_bwd_sgd_sorted,
         "Atomic-free segmented scatter+SGD over SORTED ids");
+  m.def("emb_bwd_sgd_fused_wide", &emb_bwd_sgd_fused_wide,
+        "Fused deep scatter+SGD + wide scalar scatter (shared ids)");
   m.def("emb_bwd_dense", &emb_bwd_dense,
         "Sparse embedding grad scatter into dense grad table");
   m.def("bias_relu_fwd", &bias_relu_fwd, "Fused bias+ReLU forward");

@@ -90,6 +90,40 @@ __global__ void emb_bwd_dense_kernel(
   }
 }
 
+// Fused deep+wide sparse update: the CTR models' wide (dim-1) table is
+// driven by the SAME flat ids as the deep (dim-16) table, so one kernel
+// reads the ids once and issues both updates (separate kernels re-read
+// 13.6 MB of ids and pay an extra launch; measured 84 us for the wide
+// scatter alone at b=65536).  The lane covering quad 0 of each row adds
+// the wide scalar.
+template <typename GIo>
+__global__ void emb_bwd_sgd_fused_wide_kernel(
+    float* __restrict__ table, float* __restrict__ wide_table,
+    const int64_t* __restrict__ ids,
+    const typename GIo::scalar_t* __restrict__ g,
+    const typename GIo::scalar_t* __restrict__ gw,
+    int64_t n_rows, int64_t dim, int g_div,
+    float neg_lr_scale, float wide_alpha) {
+  const int64_t dvec = dim >> 2;
+  const int64_t total = n_rows * dvec;
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       t < total; t += stride) {
+    const int64_t row = t / dvec;
+    const int64_t c4 = t - row * dvec;
+    const int64_t id = ids[row];
+    float* dst = table + id * dim + c4 * 4;
+    float gv[4];
+    QuadIo<GIo>::load4(g, row * dvec + c4, gv);
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      f32_atomic_add(dst + j, neg_lr_scale * gv[j]);
+    if (c4 == 0)
+      f32_atomic_add(wide_table + id,
+                     wide_alpha * GIo::load(gw, row / g_div));
+  }
+}
+
 // Scalar variants for dim % 4 != 0 (e.g. the wide part's dim-1 tables).
 template <typename OIo>
 __global__ void emb_fwd_scalar_kernel(const float* __restrict__ table,
@@ -492,5 +526,45 @@ void emb_bwd_sgd_sorted(torch::Tensor table, torch::Tensor sorted_ids,
                        sorted_ids.data_ptr<int64_t>(),
                        reinterpret_cast<unsigned short*>(grad.data_ptr()),
                        n, dim, nls);
+  }
+}
+
+void emb_bwd_sgd_fused_wide(torch::Tensor table, torch::Tensor wide_table,
+                            torch::Tensor ids, torch::Tensor grad,
+                            torch::Tensor gw, double lr, double scale) {
+  const int64_t dim = table.size(1);
+  const int64_t n = ids.numel();
+  check_emb(table, ids, dim);
+  debug_check_ids(ids, table.size(0));
+  TORCH_CHECK(dim % 4 == 0, "fused wide update needs dim % 4 == 0");
+  TORCH_CHECK(wide_table.is_cuda() && wide_table.is_contiguous() &&
+              wide_table.scalar_type() == torch::kFloat32 &&
+              wide_table.size(0) == table.size(0), "bad wide table");
+  TORCH_CHECK(grad.is_cuda() && grad.is_contiguous() &&
+              grad.numel() == n * dim, "grad shape mismatch");
+  TORCH_CHECK(gw.is_cuda() && gw.is_contiguous() &&
+              gw.scalar_type() == grad.scalar_type() &&
+              n % gw.numel() == 0, "gw dtype/shape mismatch");
+  const int g_div = static_cast<int>(n / gw.numel());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int grid = miyarn_grid_cap(n * (dim / 4), 8192);
+  const float nls = (float)(-lr * scale);
+  if (grad.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(emb_bwd_sgd_fused_wide_kernel<F32Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream,
+                       table.data_ptr<float>(),
+                       wide_table.data_ptr<float>(),
+                       ids.data_ptr<int64_t>(), grad.data_ptr<float>(),
+                       gw.data_ptr<float>(), n, dim, g_div, nls, nls);
+  } else {
+    TORCH_CHECK(grad.scalar_type() == torch::kBFloat16, "fp32/bf16 only");
+    hipLaunchKernelGGL(emb_bwd_sgd_fused_wide_kernel<Bf16Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream,
+                       table.data_ptr<float>(),
+                       wide_table.data_ptr<float>(),
+                       ids.data_ptr<int64_t>(),
+                       reinterpret_cast<unsigned short*>(grad.data_ptr()),
+                       reinterpret_cast<unsigned short*>(gw.data_ptr()),
+                       n, dim, g_div, nls, nls);
   }
 }

@@ -5,6 +5,14 @@
 // the 1024x432 layer).  256x256 tiles halve that again (~0.5 GB), the
 // regime where the B-sweep says the structure already matches hipBLASLt.
 //
+// Double-buffering post-mortem (round 2): two LDS buffer pairs with one
+// barrier per k-step were tried twice — runtime buffer select (537 TF)
+// and a 2x-unrolled constant-offset pipeline (544/580 TF) — and BOTH
+// lost to this single-buffered loop (581/655 TF measured).  At 2
+// waves/SIMD the two-barrier alternation keeps the MFMA and LDS-staging
+// phases of the co-resident waves interleaved; the "saved" barrier was
+// not the bottleneck.  Keep single-buffered.
+//
 // Geometry: 8 waves (512 threads); wave quadrants 4(N) x 2(M), each wave
 // 64x128 = 4x8 fragments of mfma_f32_16x16x32_bf16 (128 fp32 acc/lane —
 // ~2 waves/SIMD occupancy, 1 block/CU with 64 KB LDS).  Staging, LDS
@@ -49,15 +57,9 @@ void wgrad256_kernel(const unsigned short* __restrict__ dy,
   const int64_t k_begin = (int64_t)blockIdx.y * chunk;
   const int64_t k_end = min(B, k_begin + chunk);
 
-  // Double-buffered LDS (v5): 2 x (dy 32 KB + x 32 KB) = 128 KB of the
-  // 160 KB budget.  Occupancy is VGPR-bound at 1 block/CU either way
-  // (128 fp32 acc/lane), so the second buffer is free — and it cuts the
-  // k-loop from two barriers per step to ONE: MFMAs consume buffer A
-  // while the next tile's staged registers store into buffer B.
-  __shared__ __attribute__((aligned(16)))
-      unsigned char lds_raw[2 * 2 * 256 * 128];
-  constexpr int kBufBytes = 2 * 256 * 128;
-  // dyT(b) = [256 n][64 k] swizzled; xT(b) = [256 m][64 k]
+  __shared__ __attribute__((aligned(16))) unsigned char lds_raw[2 * 256 * 128];
+  unsigned char* dyT = lds_raw;             // [256 n][64 k] swizzled
+  unsigned char* xT = lds_raw + 256 * 128;  // [256 m][64 k]
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -100,9 +102,7 @@ void wgrad256_kernel(const unsigned short* __restrict__ dy,
     }
   };
 
-  auto write_step = [&](int buf) {
-    unsigned char* dyT = lds_raw + buf * kBufBytes;
-    unsigned char* xT = dyT + 256 * 128;
+  auto write_step = [&] {
 #pragma unroll
     for (int h = 0; h < 2; ++h)
 #pragma unroll
@@ -117,7 +117,12 @@ void wgrad256_kernel(const unsigned short* __restrict__ dy,
       }
   };
 
-  auto mfma_step = [&](const unsigned char* dyT, const unsigned char* xT) {
+  load_step(k_begin);
+  for (int64_t k0 = k_begin; k0 < k_end; k0 += W4_BK) {
+    write_step();
+    __syncthreads();
+    if (k0 + W4_BK < k_end)
+      load_step(k0 + W4_BK);  // in flight under the MFMA phase
 #pragma unroll
     for (int kh = 0; kh < 2; ++kh) {
       bf16x8_v4 a[4], b[8];
@@ -136,37 +141,7 @@ void wgrad256_kernel(const unsigned short* __restrict__ dy,
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a[i], b[j], acc[i][j], 0, 0, 0);
     }
-  };
-
-  // Pipeline, unrolled x2 so every LDS base is a compile-time constant
-  // (a runtime buf*kBufBytes select costs VALU address math on every
-  // ds access — measured 537 vs 655 TF): iteration A MFMAs buffer 0
-  // while the freshly loaded registers for the next tile store into
-  // buffer 1, ONE barrier publishes them, and B mirrors with the
-  // buffers swapped.
-  load_step(k_begin);
-  write_step(0);
-  __syncthreads();
-  int64_t k0 = k_begin;
-  while (true) {
-    bool has_next = (k0 + W4_BK) < k_end;
-    if (has_next)
-      load_step(k0 + W4_BK);  // in flight under the MFMA phase
-    mfma_step(lds_raw, lds_raw + 256 * 128);
-    if (has_next)
-      write_step(1);
     __syncthreads();
-    if (!has_next) break;
-    k0 += W4_BK;
-    has_next = (k0 + W4_BK) < k_end;
-    if (has_next)
-      load_step(k0 + W4_BK);
-    mfma_step(lds_raw + kBufBytes, lds_raw + kBufBytes + 256 * 128);
-    if (has_next)
-      write_step(0);
-    __syncthreads();
-    if (!has_next) break;
-    k0 += W4_BK;
   }
 
   float* out = part + (int64_t)blockIdx.y * N * M;
