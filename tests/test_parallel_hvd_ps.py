@@ -108,3 +108,69 @@ def test_allreduce_tensors_single_process_noop():
     ref = t.clone()
     allreduce_tensors([t])
     assert torch.equal(t, ref)
+
+
+def _hvd_grads_worker(rank, world_size, kv_addr, out_q):
+    from tf_yarn_amd.parallel import comm
+    from tf_yarn_amd.parallel.hvd import DistributedOptimizer
+    client = KVClient(kv_addr)
+    comm.init_process_group(rank=rank, world_size=world_size,
+                            backend="gloo", kv_client=client)
+    try:
+        torch.manual_seed(11)  # same init everywhere
+        model = nn.Sequential(nn.Linear(6, 8), nn.ReLU(), nn.Linear(8, 3))
+        opt = DistributedOptimizer(
+            torch.optim.SGD(model.parameters(), lr=0.1))
+        grads_out = []
+        for step in range(2):
+            torch.manual_seed(70 + step * world_size + rank)
+            x, y = torch.randn(4, 6), torch.randn(4, 3)
+            opt.zero_grad(set_to_none=False)
+            nn.functional.mse_loss(model(x), y).backward()
+            opt.synchronize()  # drains the async bucket works
+            grads_out.append([p.grad.numpy().copy()
+                              for p in model.parameters()])
+            opt.optimizer.step()  # inner step (sync already done)
+        out_q.put((rank, grads_out))
+    finally:
+        comm.destroy_process_group()
+
+
+def test_hvd_reducer_grads_match_reference():
+    """The hook-driven bucket reducer's post-synchronize grads equal the
+    mean of the per-rank reference grads (VERDICT r1 #6 done-criterion)."""
+    world_size = 2
+    server = KVServer()
+    ctx = mp.get_context("spawn")
+    out_q = ctx.Queue()
+    procs = [ctx.Process(target=_hvd_grads_worker,
+                         args=(r, world_size, server.address, out_q))
+             for r in range(world_size)]
+    for p in procs:
+        p.start()
+    try:
+        results = dict(out_q.get(timeout=120) for _ in range(world_size))
+    finally:
+        for p in procs:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.terminate()
+        server.stop()
+    # reference: same model stepping on the averaged grads
+    torch.manual_seed(11)
+    ref_model = nn.Sequential(nn.Linear(6, 8), nn.ReLU(), nn.Linear(8, 3))
+    ref_opt = torch.optim.SGD(ref_model.parameters(), lr=0.1)
+    for step in range(2):
+        ref_opt.zero_grad()
+        losses = []
+        for rank in range(world_size):
+            torch.manual_seed(70 + step * world_size + rank)
+            x, y = torch.randn(4, 6), torch.randn(4, 3)
+            losses.append(nn.functional.mse_loss(ref_model(x), y))
+        (sum(losses) / world_size).backward()
+        for rank in range(world_size):
+            for got, want in zip(results[rank][step],
+                                 ref_model.parameters()):
+                assert torch.allclose(torch.from_numpy(got), want.grad,
+                                      atol=1e-6), f"r{rank} s{step}"
+        ref_opt.step()

@@ -3,16 +3,25 @@
 Re-implements the surface the reference wires through Horovod-gloo
 (``hvd.DistributedOptimizer``,
 ``BroadcastGlobalVariablesHook/Callback(0)``, ``gloo_allred_task.py``):
-gradients are fused into flat buckets and averaged with RCCL ring
-allreduce at ``step()`` time (Horovod-gloo reduced per-tensor, unfused —
-bucketing is the xGMI-native upgrade), and rank-0 weight broadcast uses the
-same bucket buffers.
+the ``DistributedOptimizer`` wrapper averages gradients across ranks
+before ``step()``, and rank-0 weight broadcast mirrors
+``BroadcastGlobalVariables``.
+
+Round-2 redesign: gradients live as views into
+PREALLOCATED flat buckets, and per-parameter post-accumulate hooks
+launch each bucket's async allreduce as soon as its last grad lands —
+overlapped with the rest of backward, exactly the reducer technique
+proven in ``parallel/ddp.py`` — instead of the round-1 synchronous
+``torch.cat`` + copy-back inside ``step()`` (two extra full-gradient
+memory passes per step, zero overlap; Horovod's own background fusion
+thread is the reference behavior being matched).  ``step()`` just
+drains the in-flight works.
 """
 
 from __future__ import annotations
 
 import logging
-from typing import List
+from typing import List, Optional
 
 import torch
 import torch.distributed as dist
@@ -53,7 +62,8 @@ def _bucketize(tensors: List[torch.Tensor],
 def allreduce_tensors(tensors: List[torch.Tensor], average: bool = True,
                       process_group=None,
                       bucket_bytes: int = DEFAULT_BUCKET_BYTES) -> None:
-    """Fused-bucket allreduce of a tensor list (in place)."""
+    """Fused-bucket allreduce of a tensor list (in place).  One-shot
+    API (no preallocated state) for callers outside the optimizer."""
     world = _world(process_group)
     if world <= 1 or not tensors:
         return
@@ -90,9 +100,34 @@ def broadcast_parameters(module_or_params, root_rank: int = 0,
             offset += t.numel()
 
 
+class _Bucket:
+    __slots__ = ("index", "params", "buffer", "views", "pending", "seen",
+                 "work")
+
+    def __init__(self, index: int, params: List[torch.nn.Parameter]):
+        self.index = index
+        self.params = params
+        p0 = params[0]
+        numel = sum(p.numel() for p in params)
+        self.buffer = torch.zeros(numel, dtype=p0.dtype, device=p0.device)
+        self.views: List[torch.Tensor] = []
+        offset = 0
+        for p in params:
+            self.views.append(
+                self.buffer[offset:offset + p.numel()].view_as(p))
+            offset += p.numel()
+        self.pending = len(params)
+        self.seen = [False] * len(params)
+        self.work: Optional[dist.Work] = None
+
+
 class DistributedOptimizer(Optimizer):
-    """Wrap any torch optimizer with a pre-step fused-bucket gradient
-    allreduce (the ``hvd.DistributedOptimizer`` surface)."""
+    """Wrap any torch optimizer with bucketed gradient allreduce
+    overlapped with backward (the ``hvd.DistributedOptimizer``
+    surface).  Gradients become views into preallocated flat buckets;
+    each bucket's async allreduce launches (in strict bucket order, the
+    RCCL launch-order contract) when its last grad arrives, and
+    ``step()`` waits for the in-flight works before applying."""
 
     def __init__(self, optimizer: Optimizer, process_group=None,
                  average: bool = True,
@@ -105,15 +140,100 @@ class DistributedOptimizer(Optimizer):
         self.param_groups = optimizer.param_groups
         self.state = optimizer.state
         self.defaults = optimizer.defaults
+        self._world_size = _world(process_group)
+        self._buckets: List[_Bucket] = []
+        self._param_to_bucket = {}
+        self._ready: List[bool] = []
+        self._next_launch = 0
+        self._hook_handles = []
+        if self._world_size > 1:
+            self._build_buckets()
+            self._register_hooks()
 
+    # -- construction ------------------------------------------------------
+
+    def _dense_params(self):
+        return [p for g in self.optimizer.param_groups for p in g["params"]
+                if p.requires_grad
+                and not getattr(p, "_miyarn_sparse", False)]
+
+    def _build_buckets(self) -> None:
+        params = list(reversed(self._dense_params()))
+        for i, group in enumerate(_bucketize(params, self.bucket_bytes)):
+            bucket = _Bucket(i, group)
+            self._buckets.append(bucket)
+            for slot, p in enumerate(group):
+                self._param_to_bucket[p] = (bucket, slot)
+                p.grad = bucket.views[slot]
+        self._ready = [False] * len(self._buckets)
+        logger.info("hvd reducer: %d params in %d buckets",
+                    len(params), len(self._buckets))
+
+    def _register_hooks(self) -> None:
+        for p in self._param_to_bucket:
+            self._hook_handles.append(
+                p.register_post_accumulate_grad_hook(self._on_grad_ready))
+
+    # -- backward machinery ------------------------------------------------
+
+    def _on_grad_ready(self, param) -> None:
+        bucket, slot = self._param_to_bucket[param]
+        if param.grad is None:
+            return
+        if param.grad.data_ptr() != bucket.views[slot].data_ptr():
+            bucket.views[slot].copy_(param.grad)
+            param.grad = bucket.views[slot]
+        if not bucket.seen[slot]:
+            bucket.seen[slot] = True
+            bucket.pending -= 1
+        if bucket.pending == 0:
+            self._ready[bucket.index] = True
+            while (self._next_launch < len(self._buckets)
+                   and self._ready[self._next_launch]):
+                self._launch(self._buckets[self._next_launch])
+                self._next_launch += 1
+
+    def _launch(self, bucket: _Bucket) -> None:
+        op = dist.ReduceOp.AVG if (
+            self.average
+            and dist.get_backend(self.process_group) == "nccl"
+        ) else dist.ReduceOp.SUM
+        bucket.work = dist.all_reduce(bucket.buffer, op=op, async_op=True,
+                                      group=self.process_group)
+
+    @torch.no_grad()
     def synchronize(self) -> None:
-        grads = [p.grad for g in self.optimizer.param_groups
-                 for p in g["params"]
-                 if p.grad is not None
-                 and not getattr(p, "_miyarn_sparse", False)]
-        allreduce_tensors(grads, average=self.average,
-                          process_group=self.process_group,
-                          bucket_bytes=self.bucket_bytes)
+        """Drain: launch any not-yet-launched buckets (zero-filling
+        slots whose grads never arrived, matching hvd's skip-None
+        semantics) and wait for all in-flight works."""
+        if self._world_size <= 1:
+            return
+        for bucket in self._buckets[self._next_launch:]:
+            for slot, p in enumerate(bucket.params):
+                if bucket.seen[slot]:
+                    continue
+                if p.grad is not None and \
+                        p.grad.data_ptr() != bucket.views[slot].data_ptr():
+                    bucket.views[slot].copy_(p.grad)
+                    p.grad = bucket.views[slot]
+                elif p.grad is None:
+                    bucket.views[slot].zero_()
+                    p.grad = bucket.views[slot]
+        while self._next_launch < len(self._buckets):
+            self._launch(self._buckets[self._next_launch])
+            self._next_launch += 1
+        scale_needed = (self.average and
+                        dist.get_backend(self.process_group) != "nccl")
+        for bucket in self._buckets:
+            if bucket.work is not None:
+                bucket.work.wait()
+                bucket.work = None
+            if scale_needed:
+                bucket.buffer.div_(self._world_size)
+            bucket.pending = len(bucket.params)
+            bucket.seen = [False] * len(bucket.params)
+        self._ready = [False] * len(self._buckets)
+        self._next_launch = 0
 
     def step(self, closure=None):
         self.synchronize()
