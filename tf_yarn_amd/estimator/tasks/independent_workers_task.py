@@ -93,9 +93,18 @@ def main() -> None:
         return
 
     _task_commons.choose_master(client, topo.rank)
+    # RCCL p2p pair groups need one DISTINCT GPU per rank (RCCL refuses
+    # duplicate devices in a communicator).  With fewer GPUs than PS
+    # ranks, fall back to gloo transport with CPU wire buffers while
+    # keeping shards + fused optimizer apply on the GPU (staged mode —
+    # also the reference's CPU-ps/GPU-worker deployment shape).
+    use_rccl = (device.startswith("cuda")
+                and topo.world_size <= torch.cuda.device_count())
+    backend = "nccl" if use_rccl else "gloo"
+    comm_device = None if (use_rccl or not device.startswith("cuda")) \
+        else "cpu"
     comm.init_process_group(rank=topo.rank, world_size=topo.world_size,
-                            backend="nccl" if device.startswith("cuda")
-                            else "gloo",
+                            backend=backend,
                             device=device, kv_client=client,
                             group_name="ps_world", need_subgroups=True)
     pair_groups = ps_mod.build_pair_groups(topo)
@@ -108,10 +117,12 @@ def main() -> None:
     thread = None
     try:
         if topo.is_ps:
-            _run_ps(client, topo, layout, pair_groups, device, estimator)
+            _run_ps(client, topo, layout, pair_groups, device, estimator,
+                    comm_device)
         else:
             thread = _run_training(client, topo, layout, pair_groups,
-                                   device, experiment, task_key.type)
+                                   device, experiment, task_key.type,
+                                   comm_device)
             # ps never joins in the reference (:38-40); training tasks do
             thread.join()
     finally:
@@ -135,7 +146,7 @@ def _run_local(client, experiment: Experiment, task_type: str):
 
 
 def _run_ps(client, topo, layout, pair_groups, device,
-            estimator: Estimator) -> None:
+            estimator: Estimator, comm_device=None) -> None:
     """Serve a parameter shard with the user's optimizer applied to the
     flat shard (fused HIP step on GPU)."""
     shard_index = topo.rank - topo.n_workers
@@ -148,7 +159,8 @@ def _run_ps(client, topo, layout, pair_groups, device,
         opt.step()
 
     server = ps_mod.PsShardServer(topo, layout, pair_groups, device,
-                                  optimizer_step)
+                                  optimizer_step,
+                                  comm_device=comm_device)
     server.shard = shard_param.data  # optimizer updates this in place
     server.receive_initial(src_rank=0)
     event.start_event(client, _task_commons.get_task())
@@ -158,14 +170,15 @@ def _run_ps(client, topo, layout, pair_groups, device,
 
 
 def _run_training(client, topo, layout, pair_groups, device,
-                  experiment: Experiment, task_type: str):
+                  experiment: Experiment, task_type: str,
+                  comm_device=None):
     estimator: Estimator = experiment.estimator
     torch.manual_seed(INIT_SEED)
     estimator._ensure_built()
     params = [p for p in estimator._module.parameters()
               if p.requires_grad]
     channel = ps_mod.PsWorkerChannel(topo, layout, pair_groups, device,
-                                     params)
+                                     params, comm_device=comm_device)
     if topo.rank == 0:
         channel.send_initial()
 
