@@ -14,8 +14,20 @@ import sys
 import time
 
 # MIOpen exhaustive per-conv search takes minutes on a fresh box; FAST
-# find keeps warmup within the bench budget.
+# find keeps warmup within the bench budget.  A pre-warmed user find-db
+# captured on MI355X ships under tuned/miopen (the hipBLASLt-table
+# pattern): seed a scratch copy so fresh driver boxes start tuned and
+# the repo tree stays read-only (MIOpen writes/locks its db files).
 os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+_REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+_MIOPEN_TUNED = os.path.join(_REPO, "tuned", "miopen")
+if os.path.isdir(_MIOPEN_TUNED) and "MIOPEN_USER_DB_PATH" not in os.environ:
+    import shutil
+    import tempfile
+    _mdir = tempfile.mkdtemp(prefix="miyarn_miopen_")
+    for _f in os.listdir(_MIOPEN_TUNED):
+        shutil.copy(os.path.join(_MIOPEN_TUNED, _f), _mdir)
+    os.environ["MIOPEN_USER_DB_PATH"] = _mdir
 
 import torch
 import torch.distributed as dist
