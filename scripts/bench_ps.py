@@ -58,18 +58,31 @@ def main() -> None:
     ap.add_argument("--batch", type=int, default=2048)
     args = ap.parse_args()
 
+    import tempfile
+
     from tf_yarn_amd import TaskSpec
+    from tf_yarn_amd.topologies import NodeLabel
     from tf_yarn_amd.estimator import run_on_yarn
 
-    model_dir = os.environ.get("MODEL_DIR", "/tmp/miyarn_bench_ps")
+    # FRESH model dir: a stale one makes estimator.train no-op at
+    # global_step >= max_steps and the bench reports garbage.
+    model_dir = os.environ.get("MODEL_DIR") or tempfile.mkdtemp(
+        prefix="miyarn_bench_ps_")
+    # GPU-label the PS world when the box has a GPU (config 2 places all
+    # four tasks on MI355X; a 1-GPU box timeshares cuda:0).
+    label = (NodeLabel.GPU if torch.cuda.is_available()
+             and os.environ.get("MIYARN_PS_CPU", "") != "1"
+             else NodeLabel.CPU)
     n_train_tasks = 3  # chief + 2 workers train concurrently (async PS)
     t0 = time.perf_counter()
     metrics = run_on_yarn(
         experiment_fn_factory(model_dir, args.steps, args.batch),
         {
-            "chief": TaskSpec(memory=2048, vcores=2),
-            "ps": TaskSpec(memory=2048, vcores=2, instances=1),
-            "worker": TaskSpec(memory=2048, vcores=2, instances=2),
+            "chief": TaskSpec(memory=2048, vcores=8, label=label),
+            "ps": TaskSpec(memory=2048, vcores=8, instances=1,
+                           label=label),
+            "worker": TaskSpec(memory=2048, vcores=8, instances=2,
+                               label=label),
         })
     wall = time.perf_counter() - t0
     train_s = metrics.total_training_duration or wall

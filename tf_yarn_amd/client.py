@@ -114,6 +114,16 @@ def _allocate_gpus(tasks: List[ContainerTask],
     assignment is static — contiguous GPU ids per container keep each
     container's processes on xGMI-adjacent devices.
     """
+    # Clamp the node model (NODE_GPU_COUNT) to what the machine really
+    # has: on a 1-GPU box a 4-task GPU topology would otherwise hand out
+    # cuda:1..3 (nonexistent devices).
+    n_gpus = constants.NODE_GPU_COUNT
+    try:
+        import torch
+        if torch.cuda.is_available():
+            n_gpus = max(1, torch.cuda.device_count())
+    except Exception:  # noqa: BLE001 - torch-free client stays usable
+        pass
     assignment: Dict[str, List[int]] = {}
     next_gpu = 0
     for t in tasks:
@@ -121,9 +131,8 @@ def _allocate_gpus(tasks: List[ContainerTask],
         key = f"{t.type}:{t.id}"
         if (spec is not None and spec.label == NodeLabel.GPU
                 and t.type in ("chief", "worker", "ps")):
-            gpus = [(next_gpu + i) % constants.NODE_GPU_COUNT
-                    for i in range(t.nb_proc)]
-            next_gpu = (next_gpu + t.nb_proc) % constants.NODE_GPU_COUNT
+            gpus = [(next_gpu + i) % n_gpus for i in range(t.nb_proc)]
+            next_gpu = (next_gpu + t.nb_proc) % n_gpus
             assignment[key] = gpus
         else:
             assignment[key] = []
@@ -158,6 +167,12 @@ def _task_env(task: ContainerTask,
     child[constants.ENV_N_TRY] = str(n_try)
     child["MIYARN_GPU_IDS"] = ",".join(str(g) for g in gpus)
     child["MIYARN_LOG_FILE"] = log_path
+    # vcores bound the container's CPU parallelism (the reference's YARN
+    # vcore semantics, topologies.py:8-9) — and prevent N tasks x
+    # all-cores OMP oversubscription on small hosts.
+    spec = task_specs.get(task.type)
+    if spec is not None:
+        child.setdefault("OMP_NUM_THREADS", str(max(1, spec.vcores)))
     # RCCL over xGMI: dmabuf IPC only on this driver.
     child.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     if mlflow.use_mlflow:
