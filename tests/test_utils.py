@@ -130,3 +130,84 @@ def test_analyze_rocpd_tool(tmp_path):
         cwd=repo, capture_output=True, text=True, timeout=60)
     assert out.returncode == 0, out.stderr
     assert "k1" in out.stdout and "us/step" in out.stdout
+
+
+# ---- round-2 additions -----------------------------------------------------
+
+def test_commprobe_disabled_is_noop_and_enabled_records():
+    from tf_yarn_amd.utils import commprobe
+    commprobe.reset()
+    commprobe.disable()
+    with commprobe.span("x"):
+        pass
+    commprobe.add_host_time("y", 1.0)
+    assert commprobe.summary() == {}
+    commprobe.enable()
+    with commprobe.span("x"):
+        pass
+    commprobe.add_host_time("y", 0.25)
+    s = commprobe.summary()
+    commprobe.disable()
+    assert s["y"]["ms"] == 250.0 and s["y"]["count"] == 1
+    assert s["x"]["count"] == 1 and s["x"]["ms"] >= 0.0
+    commprobe.reset()
+
+
+def test_pick_region_bits_heuristic_bounds():
+    from tf_yarn_amd import ops
+    # bench shape: 26M rows, 1.7M updates -> mid-range bits
+    assert 10 <= ops.pick_region_bits(26_000_000, 1_700_000) <= 13
+    assert ops.pick_region_bits(100, 1_000_000) == 7      # tiny table
+    assert ops.pick_region_bits(1 << 30, 10) == 14        # huge/sparse
+    assert ops.pick_region_bits(1000, 0) == 11            # degenerate
+
+
+def test_emb_bwd_sgd_fused_wide_cpu_reference():
+    import torch
+    from tf_yarn_amd import ops
+    torch.manual_seed(3)
+    rows, batch, fan, dim = 500, 40, 5, 8
+    table = torch.randn(rows, dim)
+    wide = torch.randn(rows, 1)
+    rt, rw = table.clone(), wide.clone()
+    ids = torch.randint(0, rows, (batch * fan,))
+    grad = torch.randn(batch * fan, dim)
+    gw = torch.randn(batch)
+    ops.emb_bwd_sgd_fused_wide(table, wide, ids, grad, gw, lr=0.1,
+                               scale=0.5)
+    rt.index_add_(0, ids, grad, alpha=-0.05)
+    exp = gw.reshape(-1, 1).expand(-1, fan).reshape(-1)
+    rw.reshape(-1).index_add_(0, ids, exp, alpha=-0.05)
+    assert torch.allclose(table, rt, atol=1e-5)
+    assert torch.allclose(wide, rw, atol=1e-5)
+
+
+def test_binned_wrappers_cpu_fallback():
+    import torch
+    from tf_yarn_amd import ops
+    torch.manual_seed(4)
+    rows, n, dim = 300, 50, 16
+    table = torch.randn(rows, dim)
+    ref = table.clone()
+    ids = torch.randint(0, rows, (n,))
+    grad = torch.randn(n, dim)
+    ops.emb_bwd_sgd_binned(table, ids, grad, lr=0.2, scale=1.0)
+    ref.index_add_(0, ids, grad, alpha=-0.2)
+    assert torch.allclose(table, ref, atol=1e-5)
+
+
+def test_bench_master_port_rank0_picks_free_port(tmp_path, monkeypatch):
+    import socket
+    import bench
+    monkeypatch.setenv("MIYARN_RDV_TAG", "utest")
+    monkeypatch.delenv("MASTER_PORT", raising=False)
+    # occupy 29500 so the fallback must move past it
+    blocker = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+    blocker.bind(("127.0.0.1", 29500))
+    try:
+        bench._ensure_master_port(0)
+        port = int(__import__("os").environ["MASTER_PORT"])
+        assert 29500 < port < 29600
+    finally:
+        blocker.close()
+        __import__("os").environ.pop("MASTER_PORT", None)
