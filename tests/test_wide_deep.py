@@ -115,3 +115,75 @@ def test_forward_with_labels_matches_separate_loss():
         fused.backward()  # grads flow through all three parts
         assert model.wide_dense.weight.grad is not None
         assert model.head.weight.grad is not None
+
+
+def test_keras_fit_trains_sink_based_embeddings():
+    """The Keras fit loop must apply the sink-based sparse updates (CTR
+    embedding grads never land in p.grad) and leave the sink drained."""
+    import torch
+
+    from tf_yarn_amd.estimator.keras import KerasModel
+    from tf_yarn_amd.models.wide_deep import FlatInputWideAndDeep
+
+    torch.manual_seed(0)
+    model = KerasModel(FlatInputWideAndDeep(
+        dense_dim=4, table_sizes=[50] * 3, embedding_dim=8,
+        hidden=(16,)))
+    model.compile(optimizer="sgd", loss="binary_crossentropy")
+    net = model.module.net
+    before = net.deep_embedding.weight.detach().clone()
+    wide_before = net.wide_embedding.weight.detach().clone()
+    g = torch.Generator().manual_seed(3)
+    x = torch.cat([torch.randn(64, 4, generator=g),
+                   torch.randint(0, 50, (64, 3), generator=g).float()],
+                  dim=1)
+    y = (torch.rand(64, generator=g) < 0.5).float()
+    model.fit(x, y, epochs=1, batch_size=16)
+    assert not torch.equal(net.deep_embedding.weight, before), \
+        "deep embedding table did not train through Keras fit"
+    assert not torch.equal(net.wide_embedding.weight, wide_before), \
+        "wide table did not train through Keras fit"
+    assert not net.deep_embedding.pending_grads(), "sink leaked"
+    assert not net.wide_embedding.pending_grads(), "sink leaked"
+
+
+def test_estimator_train_applies_sparse_updates(tmp_path):
+    """Same contract through the Estimator train loop."""
+    import torch
+
+    from tf_yarn_amd.estimator import RunConfig
+    from tf_yarn_amd.estimator.estimator import Estimator
+    from tf_yarn_amd.models.wide_deep import FlatInputWideAndDeep
+
+    torch.manual_seed(0)
+
+    def module_fn():
+        return FlatInputWideAndDeep(dense_dim=4, table_sizes=[50] * 3,
+                                    embedding_dim=8, hidden=(16,))
+
+    est = Estimator(
+        module_fn=module_fn,
+        loss_fn=lambda o, t: torch.nn.functional
+        .binary_cross_entropy_with_logits(o.float(), t.float()),
+        optimizer_fn=lambda ps: torch.optim.SGD(
+            [p for p in ps if not getattr(p, "_miyarn_sparse", False)],
+            lr=0.1),
+        model_dir=str(tmp_path), config=RunConfig())
+
+    def input_fn():
+        g = torch.Generator().manual_seed(4)
+        for _ in range(4):
+            x = torch.cat(
+                [torch.randn(32, 4, generator=g),
+                 torch.randint(0, 50, (32, 3), generator=g).float()],
+                dim=1)
+            y = (torch.rand(32, generator=g) < 0.5).float()
+            yield x, y
+
+    est._ensure_built()
+    net = est._module.net
+    before = net.deep_embedding.weight.detach().clone()
+    est.train(input_fn, steps=4, save_checkpoints=False)
+    assert not torch.equal(net.deep_embedding.weight, before), \
+        "embedding table did not train through Estimator.train"
+    assert not net.deep_embedding.pending_grads(), "sink leaked"

@@ -230,6 +230,10 @@ class Estimator:
                     if self.grad_sync_hook is not None:
                         self.grad_sync_hook(module)
                     optimizer.step()
+                    if hasattr(module, "apply_sparse_updates"):
+                        # sink-based embedding grads (CTR models)
+                        module.apply_sparse_updates(
+                            optimizer.param_groups[0].get("lr", 0.0))
                 self.global_step += 1
                 lval = float(loss.detach())
                 if self.global_step % self.config.log_step_count_steps == 0:

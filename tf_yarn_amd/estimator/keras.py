@@ -134,6 +134,7 @@ class KerasModel:
                 loss = self.loss_fn(out, by)
                 loss.backward()
                 self.optimizer.step()
+                self._apply_sparse()
                 loss_val = float(loss.detach())
                 epoch_loss += loss_val
                 n_b += 1
@@ -152,6 +153,16 @@ class KerasModel:
             for cb in callbacks:
                 cb.on_epoch_end(epoch, logs, self)
         return history
+
+    def _apply_sparse(self) -> None:
+        """CTR models (WideAndDeep) route embedding grads through a
+        sink instead of p.grad; apply the fused sparse update with the
+        optimizer's lr (and clear the sink so eval passes don't leak).
+        No-op for ordinary modules."""
+        m = self.module
+        if hasattr(m, "apply_sparse_updates"):
+            lr = self.optimizer.param_groups[0].get("lr", 0.0)
+            m.apply_sparse_updates(lr)
 
     @torch.no_grad()
     def evaluate(self, x, y, batch_size: int = 32) -> float:

@@ -429,3 +429,43 @@ class WideAndDeep(nn.Module):
             return
         self.deep_embedding.clear_pending()
         self.wide_embedding.clear_pending()
+
+
+class FlatInputWideAndDeep(nn.Module):
+    """WideAndDeep behind a single-tensor input — the shape the Keras
+    ``fit(x, y)`` / Estimator ``input_fn`` surfaces expect:
+    ``x = [dense fp32 columns | categorical id columns]``.  Importable
+    (not example-local) so whole-model Keras checkpoints pickle."""
+
+    def __init__(self, dense_dim: int = CRITEO_DENSE,
+                 table_sizes: Optional[List[int]] = None,
+                 embedding_dim: int = 16,
+                 hidden: Tuple[int, ...] = (1024, 512, 256),
+                 compute_dtype: torch.dtype = torch.float32,
+                 sharded: bool = False, process_group=None):
+        super().__init__()
+        self.dense_dim = dense_dim
+        self.net = WideAndDeep(dense_dim=dense_dim,
+                               table_sizes=table_sizes,
+                               embedding_dim=embedding_dim, hidden=hidden,
+                               compute_dtype=compute_dtype,
+                               sharded=sharded,
+                               process_group=process_group)
+
+    def forward(self, x: torch.Tensor,
+                labels: Optional[torch.Tensor] = None) -> torch.Tensor:
+        dense = x[:, :self.dense_dim]
+        ids = x[:, self.dense_dim:].long()
+        return self.net(dense, ids, labels=labels)
+
+    def start_sparse_sync(self, process_group=None) -> None:
+        self.net.start_sparse_sync(process_group)
+
+    def finish_sparse_sync(self, lr: float, stream=None) -> None:
+        self.net.finish_sparse_sync(lr, stream=stream)
+
+    def apply_sparse_updates(self, lr: float, process_group=None) -> None:
+        self.net.apply_sparse_updates(lr, process_group)
+
+    def clear_pending(self) -> None:
+        self.net.clear_pending()
