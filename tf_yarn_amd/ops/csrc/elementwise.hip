@@ -389,11 +389,14 @@ std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
   const int block8 = octs >= 128 ? static_cast<int>(octs) : MIYARN_BLOCK;
   const int64_t rpb = oct_ok ? block8 / octs : 1;
   if (oct_ok)
-    // cap grid*rpb (= partial rows) at MAX_BLOCKS so the host-side
-    // partial reduction stays small; slots grid-stride over rows
+    // Fill the chip: narrow layers (256 cols -> rpb=8) were capped at
+    // 256 blocks (1/CU) by a MAX_BLOCKS/rpb partial-row bound and ran
+    // at 1.9 TB/s; partial rows up to 8*MAX_BLOCKS cost only a ~16 MB
+    // torch reduce (~2 us) while the streaming phase gains the other
+    // 7 blocks/CU.
     grid = static_cast<int>(std::min<int64_t>(
         (rows + rpb * DB_ROWS - 1) / (rpb * DB_ROWS),
-        MIYARN_MAX_BLOCKS / rpb));
+        MIYARN_MAX_BLOCKS));
   // Atomic-free column partials: one row per block (per slot for the
   // oct kernel), reduced below.
   auto part = torch::empty({grid * rpb, cols},
