@@ -38,10 +38,16 @@ union U16x4c {
 
 __device__ __forceinline__ int lds_off4(int n, int k) {
   int byte = n * 128 + k * 2;
-  // combined swizzle: (n&7) spreads the b128 READ groups (consecutive
-  // rows), (n>>3)&7 spreads the b64 WRITE groups (lanes stride 8 rows —
-  // with the plain (n&7) term those writes were a 16-way conflict)
-  return byte ^ (((n ^ (n >> 3)) & 7) << 4);
+  // combined swizzle (v1.5): (n&7) spreads the b128 READ groups
+  // (consecutive rows), (n>>3)&7 spreads the b64 WRITE groups (the
+  // plain (n&7) term made those a 16-way conflict), and the (n>>6)
+  // bit-7 term splits the write groups across the two 128 B halves of
+  // a bank sweep — 4-way -> 2-way write conflicts (the 16 B-granular
+  // XOR floor for this staging shape; verified by bank simulation and
+  // PMC SQ_LDS_BANK_CONFLICT).  Reads stay at the conflict-free 4
+  // (bit 7 is constant across each 16-row read group).
+  const int s = ((n ^ (n >> 3)) & 7) | (((n >> 6) & 1) << 3);
+  return byte ^ (s << 4);
 }
 
 __global__ __launch_bounds__(512)
