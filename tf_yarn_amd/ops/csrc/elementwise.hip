@@ -391,12 +391,13 @@ std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
   if (oct_ok)
     // Fill the chip: narrow layers (256 cols -> rpb=8) were capped at
     // 256 blocks (1/CU) by a MAX_BLOCKS/rpb partial-row bound and ran
-    // at 1.9 TB/s; partial rows up to 8*MAX_BLOCKS cost only a ~16 MB
-    // torch reduce (~2 us) while the streaming phase gains the other
-    // 7 blocks/CU.
+    // at 1.9 TB/s.  Allow up to 4x MAX_BLOCKS partial rows (>= 4
+    // blocks/CU for every width) — unbounded partials measured the
+    // dbias reduce growing +18 us/step, eating half the streaming win.
     grid = static_cast<int>(std::min<int64_t>(
         (rows + rpb * DB_ROWS - 1) / (rpb * DB_ROWS),
-        MIYARN_MAX_BLOCKS));
+        std::min<int64_t>(MIYARN_MAX_BLOCKS,
+                          4 * MIYARN_MAX_BLOCKS / rpb)));
   // Atomic-free column partials: one row per block (per slot for the
   // oct kernel), reduced below.
   auto part = torch::empty({grid * rpb, cols},
