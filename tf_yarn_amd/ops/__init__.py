@@ -476,9 +476,10 @@ class BiasReLU(torch.autograd.Function):
             # v3 fused kernel: atomic-free per-block column partials +
             # one tiny reduce (earlier atomic/LDS variants measured
             # slower than the separate path; see micro_elementwise.py).
-            dx, dbias32 = _C.bias_relu_bwd_db(dy.contiguous(),
-                                              y.contiguous())
-            return dx, dbias32.to(dy.dtype)
+            dx, dbias = _C.bias_relu_bwd_db(
+                dy.contiguous(), y.contiguous(),
+                dy.dtype == torch.bfloat16)
+            return dx, dbias.to(dy.dtype)
         dx = bias_relu_bwd(dy.contiguous(), y)
         dims = tuple(range(dx.dim() - 1))
         dbias = dx.sum(dim=dims)
@@ -556,15 +557,20 @@ class LinearBiasReLU(torch.autograd.Function):
         dy = dy.contiguous()
         if dy.is_cuda and HAVE_EXT and y.size(-1) % 4 == 0 \
                 and y.size(-1) <= 8192:
-            dz, dbias32 = _C.bias_relu_bwd_db(dy, y.contiguous())
-            dbias = dbias32.to(dy.dtype)
+            dz, dbias = _C.bias_relu_bwd_db(
+                dy, y.contiguous(), dy.dtype == torch.bfloat16)
+            if dbias.dtype != dy.dtype:
+                dbias = dbias.to(dy.dtype)
         else:
             dz = bias_relu_bwd(dy, y)
             dbias = dz.sum(dim=tuple(range(dz.dim() - 1)))
         dx = dz.matmul(weight)
         kernel = _custom_wgrad_kernel(dz, x)
         if kernel is not None:
-            dw = kernel(dz, x, 0).to(weight.dtype)
+            # split-K partial sum + cast fused into one reduce kernel
+            dw = kernel(dz, x, 0, weight.dtype == torch.bfloat16)
+            if dw.dtype != weight.dtype:
+                dw = dw.to(weight.dtype)
         else:
             dw = dz.t().matmul(x)
         return dx, dw, dbias

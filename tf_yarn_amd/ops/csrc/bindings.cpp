@@ -65,7 +65,9 @@ void emb_scatter_sum_binned(torch::Tensor table, torch::Tensor ids,
 torch::Tensor bias_relu_fwd(torch::Tensor x, torch::Tensor bias);
 torch::Tensor bias_relu_bwd(torch::Tensor dy, torch::Tensor y);
 std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
-                                            torch::Tensor y);
+                                            torch::Tensor y,
+                                            bool dbias_bf16);
+torch::Tensor reduce_splitk(torch::Tensor part, bool out_bf16);
 torch::Tensor col_reduce_dot(torch::Tensor x, torch::Tensor dy);
 torch::Tensor row_dot(torch::Tensor x, torch::Tensor w,
                       c10::optional<torch::Tensor> bias);
@@ -77,9 +79,12 @@ torch::Tensor bce_head_bwd(torch::Tensor sig, torch::Tensor labels,
                            torch::Tensor g);
 
 // wgrad.hip
-torch::Tensor wgrad_nt(torch::Tensor dy, torch::Tensor x, int64_t splitk);
-torch::Tensor wgrad_nt128(torch::Tensor dy, torch::Tensor x, int64_t splitk);
-torch::Tensor wgrad_nt256(torch::Tensor dy, torch::Tensor x, int64_t splitk);
+torch::Tensor wgrad_nt(torch::Tensor dy, torch::Tensor x, int64_t splitk,
+                       bool out_bf16);
+torch::Tensor wgrad_nt128(torch::Tensor dy, torch::Tensor x,
+                          int64_t splitk, bool out_bf16);
+torch::Tensor wgrad_nt256(torch::Tensor dy, torch::Tensor x,
+                          int64_t splitk, bool out_bf16);
 void convert_scaled(torch::Tensor src, torch::Tensor dst, double scale);
 
 // gemm_bt.hip
@@ -105,12 +110,21 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Sparse embedding grad scatter into dense grad table");
   m.def("bias_relu_fwd", &bias_relu_fwd, "Fused bias+ReLU forward");
   m.def("bias_relu_bwd", &bias_relu_bwd, "Fused ReLU backward");
+  m.def("reduce_splitk", &reduce_splitk,
+        "Split-K partial reduction with fused output cast",
+        py::arg("part"), py::arg("out_bf16") = false);
   m.def("wgrad_nt", &wgrad_nt,
-        "Split-K MFMA weight gradient: dW = dy^T @ x (bf16 in, fp32 out)");
+        "Split-K MFMA weight gradient: dW = dy^T @ x (bf16 in)",
+        py::arg("dy"), py::arg("x"), py::arg("splitk"),
+        py::arg("out_bf16") = false);
   m.def("wgrad_nt128", &wgrad_nt128,
-        "Split-K MFMA weight gradient, 128x128 tiles + XOR-swizzled LDS");
+        "Split-K MFMA weight gradient, 128x128 tiles + XOR-swizzled LDS",
+        py::arg("dy"), py::arg("x"), py::arg("splitk"),
+        py::arg("out_bf16") = false);
   m.def("wgrad_nt256", &wgrad_nt256,
-        "Split-K MFMA weight gradient, 256x256 tiles (8 waves)");
+        "Split-K MFMA weight gradient, 256x256 tiles (8 waves)",
+        py::arg("dy"), py::arg("x"), py::arg("splitk"),
+        py::arg("out_bf16") = false);
   m.def("col_reduce_dot", &col_reduce_dot,
         "dw[m] = sum_b dy[b] * x[b,m] (single-logit head wgrad)");
   m.def("row_dot", &row_dot,
@@ -120,7 +134,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bce_head_bwd", &bce_head_bwd,
         "dlogit = (sigmoid - label) * upstream grad");
   m.def("bias_relu_bwd_db", &bias_relu_bwd_db,
-        "Fused ReLU backward + dbias reduction (returns [dx, dbias_fp32])");
+        "Fused ReLU backward + dbias reduction (returns [dx, dbias])",
+        py::arg("dy"), py::arg("y"), py::arg("dbias_bf16") = false);
   m.def("emb_fwd_into", &emb_fwd_into,
         "Embedding gather into a slice of a larger 2D buffer");
   m.def("emb_gather_sum", &emb_gather_sum,

@@ -256,6 +256,34 @@ __global__ void f32_to_bf16_kernel(const float* __restrict__ src,
     dst[i] = f32_to_bf16(src[i] * scale);
 }
 
+
+// Split-K / column-partial reduction with the output cast fused:
+// out[j] (fp32 or bf16) = sum_s part[s][j].  Replaces torch's
+// reduce_kernel + the separate .to(bf16) elementwise pass on the wgrad
+// and dbias paths (8 reduce + 3 cast kernels, ~125 us/step at the
+// bench shape).  Threads own f32x4 quads; s-slices stream coalesced.
+template <typename OIo>
+__global__ void reduce_splitk_kernel(
+    const float* __restrict__ part,
+    typename OIo::scalar_t* __restrict__ out, int S, int64_t nm) {
+  const int64_t quads = nm >> 2;
+  const int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       q < quads; q += stride) {
+    f32x4 acc = reinterpret_cast<const f32x4*>(part)[q];
+    for (int s = 1; s < S; ++s) {
+      const f32x4 v =
+          reinterpret_cast<const f32x4*>(part + (int64_t)s * nm)[q];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[j] += v[j];
+    }
+    float vv[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) vv[j] = acc[j];
+    QuadIo<OIo>::store4(out, q, vv);
+  }
+}
+
 }  // namespace
 
 torch::Tensor bias_relu_fwd(torch::Tensor x, torch::Tensor bias) {
@@ -360,8 +388,11 @@ torch::Tensor bias_relu_bwd(torch::Tensor dy, torch::Tensor y) {
   return dx;
 }
 
+torch::Tensor reduce_splitk(torch::Tensor part, bool out_bf16);
+
 std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
-                                            torch::Tensor y) {
+                                            torch::Tensor y,
+                                            bool dbias_bf16) {
   TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && y.is_contiguous(),
               "dy/y must be GPU contiguous");
   TORCH_CHECK(dy.scalar_type() == y.scalar_type() &&
@@ -424,7 +455,13 @@ std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
                        reinterpret_cast<unsigned short*>(dx.data_ptr()),
                        part.data_ptr<float>(), rows, cols);
   }
-  auto dbias = part.sum(0);
+  torch::Tensor dbias;
+  if (cols % 4 == 0)
+    dbias = reduce_splitk(part, dbias_bf16);
+  else {
+    dbias = part.sum(0);
+    if (dbias_bf16) dbias = dbias.to(torch::kBFloat16);
+  }
   return {dx, dbias};
 }
 
@@ -547,7 +584,7 @@ torch::Tensor col_reduce_dot(torch::Tensor x, torch::Tensor dy) {
                        reinterpret_cast<unsigned short*>(dy.data_ptr()),
                        part.data_ptr<float>(), rows, cols);
   }
-  return part.sum(0);
+  return cols % 4 == 0 ? reduce_splitk(part, false) : part.sum(0);
 }
 
 namespace {
@@ -716,4 +753,31 @@ torch::Tensor bce_head_bwd(torch::Tensor sig, torch::Tensor labels,
                      labels.data_ptr<float>(), g.data_ptr<float>(),
                      reinterpret_cast<unsigned short*>(dl.data_ptr()), n);
   return dl;
+}
+
+torch::Tensor reduce_splitk(torch::Tensor part, bool out_bf16) {
+  TORCH_CHECK(part.is_cuda() && part.is_contiguous() && part.dim() >= 2 &&
+              part.scalar_type() == torch::kFloat32, "bad partials");
+  const int S = static_cast<int>(part.size(0));
+  const int64_t nm = part.numel() / S;
+  TORCH_CHECK(nm % 4 == 0, "reduce_splitk needs inner numel %% 4 == 0");
+  auto sizes = part.sizes().vec();
+  sizes.erase(sizes.begin());
+  auto out = torch::empty(sizes, part.options().dtype(
+      out_bf16 ? torch::kBFloat16 : torch::kFloat32));
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int grid = miyarn_grid(nm / 4);
+  if (out_bf16) {
+    hipLaunchKernelGGL(reduce_splitk_kernel<Bf16Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream,
+                       part.data_ptr<float>(),
+                       reinterpret_cast<unsigned short*>(out.data_ptr()),
+                       S, nm);
+  } else {
+    hipLaunchKernelGGL(reduce_splitk_kernel<F32Io>, dim3(grid),
+                       dim3(MIYARN_BLOCK), 0, stream,
+                       part.data_ptr<float>(), out.data_ptr<float>(),
+                       S, nm);
+  }
+  return out;
 }

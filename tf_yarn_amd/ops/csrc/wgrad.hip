@@ -149,8 +149,10 @@ void wgrad_nt_kernel(const unsigned short* __restrict__ dy,
 
 }  // namespace
 
+torch::Tensor reduce_splitk(torch::Tensor part, bool out_bf16);
+
 torch::Tensor wgrad_nt(torch::Tensor dy, torch::Tensor x,
-                       int64_t splitk) {
+                       int64_t splitk, bool out_bf16) {
   TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.dim() == 2 &&
               dy.scalar_type() == torch::kBFloat16,
               "dy must be [B, N] bf16 contiguous");
@@ -179,5 +181,7 @@ torch::Tensor wgrad_nt(torch::Tensor dy, torch::Tensor x,
                      reinterpret_cast<unsigned short*>(dy.data_ptr()),
                      reinterpret_cast<unsigned short*>(x.data_ptr()),
                      part.data_ptr<float>(), B, N, M, chunk);
-  return part.sum(0);
+  // fused split-K reduce (+bf16 cast when the consumer is a
+  // bf16 parameter) — replaces torch reduce + .to elementwise
+  return reduce_splitk(part, out_bf16);
 }
