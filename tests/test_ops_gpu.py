@@ -740,3 +740,28 @@ def test_emb_bwd_sgd_fused_wide_matches_separate(gdtype):
     atol = 1e-4 if gdtype == torch.float32 else 2e-2
     assert torch.allclose(table, ref_t, atol=atol)
     assert torch.allclose(wide, ref_w, atol=atol)
+
+
+@requires_gpu
+@pytest.mark.parametrize("out_bf16", [False, True])
+def test_reduce_splitk_matches_torch_sum(out_bf16):
+    import tf_yarn_amd.ops._C as C
+    torch.manual_seed(9)
+    part = torch.randn(28, 1024, 432, device="cuda")
+    got = C.reduce_splitk(part, out_bf16)
+    want = part.sum(0)
+    assert got.dtype == (torch.bfloat16 if out_bf16 else torch.float32)
+    atol = 5e-2 if out_bf16 else 1e-3
+    assert torch.allclose(got.float(), want, atol=atol, rtol=1e-2)
+
+
+@requires_gpu
+def test_wgrad_out_bf16_matches_fp32_cast():
+    import tf_yarn_amd.ops._C as C
+    torch.manual_seed(10)
+    dy = torch.randn(4096, 256, device="cuda").to(torch.bfloat16)
+    x = torch.randn(4096, 512, device="cuda").to(torch.bfloat16)
+    f32 = C.wgrad_nt256(dy, x, 8)
+    bf = C.wgrad_nt256(dy, x, 8, True)
+    assert bf.dtype == torch.bfloat16
+    assert torch.allclose(bf.float(), f32.to(torch.bfloat16).float())
