@@ -761,6 +761,15 @@ torch::Tensor reduce_splitk(torch::Tensor part, bool out_bf16) {
   const int S = static_cast<int>(part.size(0));
   const int64_t nm = part.numel() / S;
   TORCH_CHECK(nm % 4 == 0, "reduce_splitk needs inner numel %% 4 == 0");
+  if (S > 64 || nm < 4 * MIYARN_BLOCK) {
+    // The vectorized kernel parallelizes over the inner dim only; deep
+    // stacks of narrow partials (the dbias case: S up to 8192, nm 256)
+    // would serialize S-long latency chains on a handful of blocks
+    // (measured 4x whole-step regression).  torch's tree reduce is the
+    // right shape there.
+    auto out = part.sum(0);
+    return out_bf16 ? out.to(torch::kBFloat16) : out;
+  }
   auto sizes = part.sizes().vec();
   sizes.erase(sizes.begin());
   auto out = torch::empty(sizes, part.options().dtype(
