@@ -55,11 +55,13 @@ void wgrad128_kernel(const unsigned short* __restrict__ dy,
                      float* __restrict__ part,
                      int64_t B, int N, int M, int64_t chunk) {
   const int tiles_m = (M + W3_BM - 1) / W3_BM;
-  const int tile_n = blockIdx.x / tiles_m;
-  const int tile_m = blockIdx.x - tile_n * tiles_m;
+  // XCD-aware slab-major mapping (see wgrad256.hip): same-slab tiles
+  // co-locate on one XCD and share operand column reads via its L2.
+  const int tile_n = blockIdx.y / tiles_m;
+  const int tile_m = blockIdx.y - tile_n * tiles_m;
   const int n0 = tile_n * W3_BN;
   const int m0 = tile_m * W3_BM;
-  const int64_t k_begin = (int64_t)blockIdx.y * chunk;
+  const int64_t k_begin = (int64_t)blockIdx.x * chunk;
   const int64_t k_end = min(B, k_begin + chunk);
 
   __shared__ __attribute__((aligned(16))) unsigned char lds_raw[2 * 128 * 128];
@@ -156,7 +158,7 @@ void wgrad128_kernel(const unsigned short* __restrict__ dy,
   }
 
   // ---- epilogue: fp32 partial slab (column-masked) ----------------------
-  float* out = part + (int64_t)blockIdx.y * N * M;
+  float* out = part + (int64_t)blockIdx.x * N * M;
   const int c_col = lane & 15;
   const int c_row = (lane >> 4) * 4;
 #pragma unroll
@@ -202,7 +204,7 @@ torch::Tensor wgrad_nt128(torch::Tensor dy, torch::Tensor x,
   auto part = torch::empty({splitk, N, M},
                            dy.options().dtype(torch::kFloat32));
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  dim3 grid(tiles, splitk);
+  dim3 grid(splitk, tiles);
   hipLaunchKernelGGL(wgrad128_kernel, grid, dim3(256), 0, stream,
                      reinterpret_cast<unsigned short*>(dy.data_ptr()),
                      reinterpret_cast<unsigned short*>(x.data_ptr()),

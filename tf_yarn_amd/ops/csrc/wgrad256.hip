@@ -55,12 +55,19 @@ void wgrad256_kernel(const unsigned short* __restrict__ dy,
                      const unsigned short* __restrict__ x,
                      float* __restrict__ part,
                      int64_t B, int N, int M, int64_t chunk) {
+  // XCD-aware mapping: the dispatcher places linear block b on XCD
+  // b % 8 (MI355X_MICROARCH "Workgroup dispatch").  With grid =
+  // (splitk, tiles), linear id = slab + splitk*tile, so all tiles of
+  // one k-slab land on XCD slab%8 and share their dy/x column reads
+  // through that XCD's 4 MB L2 (the old (tiles, splitk) grid put every
+  // tile of a slab on a DIFFERENT XCD — zero reuse; operand re-reads
+  // made the kernel HBM-bound at ~5.6 TB/s).
   const int tiles_m = (M + W4_BM - 1) / W4_BM;
-  const int tile_n = blockIdx.x / tiles_m;
-  const int tile_m = blockIdx.x - tile_n * tiles_m;
+  const int tile_n = blockIdx.y / tiles_m;
+  const int tile_m = blockIdx.y - tile_n * tiles_m;
   const int n0 = tile_n * W4_BN;
   const int m0 = tile_m * W4_BM;
-  const int64_t k_begin = (int64_t)blockIdx.y * chunk;
+  const int64_t k_begin = (int64_t)blockIdx.x * chunk;
   const int64_t k_end = min(B, k_begin + chunk);
 
   __shared__ __attribute__((aligned(16))) unsigned char lds_raw[2 * 256 * 128];
@@ -150,7 +157,7 @@ void wgrad256_kernel(const unsigned short* __restrict__ dy,
     __syncthreads();
   }
 
-  float* out = part + (int64_t)blockIdx.y * N * M;
+  float* out = part + (int64_t)blockIdx.x * N * M;
   const int c_col = lane & 15;
   const int c_row = (lane >> 4) * 4;
 #pragma unroll
@@ -195,7 +202,7 @@ torch::Tensor wgrad_nt256(torch::Tensor dy, torch::Tensor x,
   auto part = torch::empty({splitk, N, M},
                            dy.options().dtype(torch::kFloat32));
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  dim3 grid(tiles, splitk);
+  dim3 grid(splitk, tiles);
   hipLaunchKernelGGL(wgrad256_kernel, grid, dim3(512), 0, stream,
                      reinterpret_cast<unsigned short*>(dy.data_ptr()),
                      reinterpret_cast<unsigned short*>(x.data_ptr()),
