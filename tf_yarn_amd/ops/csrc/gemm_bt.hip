@@ -39,8 +39,21 @@ void gemm_bt_kernel(const unsigned short* __restrict__ A,
                     unsigned short* __restrict__ C,
                     int64_t M, int N, int K) {
   const int tiles_n = (N + G_BN - 1) / G_BN;
-  const int tile_m = blockIdx.x / tiles_n;
-  const int tile_n = blockIdx.x - tile_m * tiles_n;
+  const int tiles_m = static_cast<int>(M / G_BM);
+  int tile_m, tile_n;
+  if ((tiles_m & 7) == 0) {
+    // XCD-aware grouping (see wgrad256.hip): linear block b lands on
+    // XCD b%8, so map the tiles_n column-tiles of each row-block to
+    // indices sharing b%8 — they re-read the SAME 256 A rows and now
+    // share them through one XCD's L2 instead of 4 different XCDs.
+    const int c = blockIdx.x & 7;
+    const int q = blockIdx.x >> 3;
+    tile_n = q % tiles_n;
+    tile_m = c + 8 * (q / tiles_n);
+  } else {
+    tile_m = blockIdx.x / tiles_n;
+    tile_n = blockIdx.x - tile_m * tiles_n;
+  }
   const int64_t m0 = (int64_t)tile_m * G_BM;
   const int n0 = tile_n * G_BN;
 
