@@ -10,6 +10,7 @@ Run: ``python -m tf_yarn_amd.bin.check_env``
 from __future__ import annotations
 
 import logging
+import os
 import sys
 
 logger = logging.getLogger(__name__)
@@ -30,6 +31,22 @@ def check_local_env() -> dict:
     results["gloo_backend"] = dist.is_gloo_available()
     from tf_yarn_amd import ops
     results["hip_extension"] = ops.HAVE_EXT
+    if ops.HAVE_EXT:
+        from tf_yarn_amd.ops import _C
+        results["kernels"] = sorted(
+            n for n in dir(_C) if not n.startswith("_"))
+    import tf_yarn_amd
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(
+        tf_yarn_amd.__file__)))
+    results["tuned_gemm_table"] = os.path.exists(
+        os.path.join(repo, "tuned", "tunableop_wide_deep.csv"))
+    results["tuned_miopen_db"] = os.path.isdir(
+        os.path.join(repo, "tuned", "miopen"))
+    try:
+        from tf_yarn_amd import _kv_native  # noqa: F401
+        results["kv_native"] = True
+    except ImportError:
+        results["kv_native"] = False
     return results
 
 
