@@ -68,3 +68,34 @@ def test_distributed_flavor_multiproc_per_worker(tmp_path):
         base_dir=str(tmp_path / "app"),
     )
     assert metrics is not None  # 4-rank allreduce inside asserted 1+2+3+4
+
+
+@pytest.mark.timeout(180)
+def test_hard_crash_kills_hung_survivors(tmp_path):
+    """A task dying WITHOUT publishing a stop event (SIGKILL/OOM shape:
+    os._exit) while its peer blocks forever on a collective must fail
+    the run in bounded time with the survivor killed — the realistic
+    GPU-box failure the event protocol cannot see (reference fail-fast
+    model: containers never restart, client.py:233)."""
+    import time as _time
+    from tf_yarn_amd import RunFailed, TaskSpec
+    from tf_yarn_amd.distributed import run_on_yarn
+
+    def crashing_fn():
+        def fn(task_params):
+            import os
+            import time
+            if task_params.rank == 0:
+                os._exit(137)  # hard death: no stop event, no traceback
+            time.sleep(3600)  # survivor hangs (e.g. stuck collective)
+        return fn
+
+    t0 = _time.perf_counter()
+    with pytest.raises(RunFailed):
+        run_on_yarn(
+            crashing_fn(),
+            {"worker": TaskSpec(memory=512, vcores=1, instances=2)},
+            base_dir=str(tmp_path / "app"),
+        )
+    # bounded: poll period + shutdown grace, nowhere near the 3600 s hang
+    assert _time.perf_counter() - t0 < 120
