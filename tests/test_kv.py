@@ -164,3 +164,32 @@ def test_add_and_cas(kv_client):
     assert kv_client.compare_set("k", b"", b"a") == b"a"
     assert kv_client.compare_set("k", b"a", b"b") == b"b"
     assert kv_client.compare_set("k", b"a", b"c") == b"b"
+
+
+def test_wait_fails_fast_when_server_dies():
+    """A blocking wait must raise (not hang) when the control-plane
+    server goes away mid-wait — tasks then fail fast instead of
+    deadlocking the whole run."""
+    import threading
+    import time
+
+    from tf_yarn_amd.kv import KVClient, KVServer
+
+    server = KVServer()
+    client = KVClient(server.address)
+    result = {}
+
+    def waiter():
+        try:
+            client.wait("never/published", timeout=60)
+            result["outcome"] = "returned"
+        except Exception as e:  # noqa: BLE001
+            result["outcome"] = f"raised {type(e).__name__}"
+
+    t = threading.Thread(target=waiter, daemon=True)
+    t.start()
+    time.sleep(0.3)
+    server.stop()
+    t.join(timeout=15)
+    assert not t.is_alive(), "wait hung after server death"
+    assert result["outcome"].startswith("raised"), result
