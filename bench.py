@@ -239,11 +239,15 @@ def main() -> None:
         dist.barrier()
     elapsed = time.perf_counter() - t0
 
-    # MAX over ranks
+    # MAX over ranks + spread (a straggling rank shows up here)
     t = torch.tensor([elapsed], dtype=torch.float64,
                      device=device if backend == "nccl" else "cpu")
+    tmin = t.clone()
     if world_size > 1:
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        dist.all_reduce(tmin, op=dist.ReduceOp.MIN)
+    rank_spread_ms = (float(t.item()) - float(tmin.item())) \
+        / args.steps * 1000.0
     elapsed = float(t.item())
 
     # -- per-phase comm/compute breakdown (extra instrumented steps AFTER
@@ -312,6 +316,11 @@ def main() -> None:
                 "parallelism": f"dp{world_size}",
                 "final_loss": float(last_loss.item())
                 if last_loss is not None else None,
+                "alltoall_mode": __import__(
+                    "tf_yarn_amd.models.sharded_embedding",
+                    fromlist=["negotiated_alltoall_mode"]
+                ).negotiated_alltoall_mode(),
+                "rank_spread_ms_per_step": rank_spread_ms,
             },
             "comm_breakdown": breakdown,
         }

@@ -58,6 +58,19 @@ def _binned_enabled() -> bool:
     return v != "" and v != "0"
 
 
+def negotiated_alltoall_mode(group=None) -> str:
+    """The mode the self-check picked for this group: 'native',
+    'emulate', 'gloo-emulate', or 'unprobed' (reported in bench.py's
+    JSON so a silent fallback is visible in the scale run)."""
+    if not (dist.is_available() and dist.is_initialized()):
+        return "unprobed"
+    if dist.get_backend(group) == "gloo":
+        return "gloo-emulate"
+    key = (group if group is not None
+           else dist.distributed_c10d._get_default_group())
+    return _ALLTOALL_MODE.get(key, "unprobed")
+
+
 def _perm_stream_enabled() -> bool:
     import os
     v = os.environ.get("MIYARN_PERM_STREAM", "1")
