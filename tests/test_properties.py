@@ -76,3 +76,78 @@ def test_feature_shard_ownership_partitions_features():
         assert flat == list(range(26))
         lens = [len(p) for p in feats]
         assert max(lens) - min(lens) <= 1
+
+
+@settings(max_examples=60, deadline=None)
+@given(rows=st.integers(1, 1 << 30), updates=st.integers(0, 1 << 30))
+def test_pick_region_bits_in_range(rows, updates):
+    from tf_yarn_amd import ops
+    bits = ops.pick_region_bits(rows, updates)
+    assert 7 <= bits <= 14
+
+
+@settings(max_examples=40, deadline=None)
+@given(n=st.integers(1, 500), bits=st.integers(1, 16),
+       rows=st.integers(1, 100_000), seed=st.integers(0, 10_000))
+def test_binned_permutation_spec_properties(n, bits, rows, seed):
+    """The executable spec's permutation invariants hold for arbitrary
+    shapes (the GPU kernel is tested against this same spec)."""
+    import torch
+
+    from tests.test_binned_scatter_spec import binned_permutation
+
+    g = torch.Generator().manual_seed(seed)
+    ids = torch.randint(0, rows, (n,), generator=g)
+    import tests.test_binned_scatter_spec as spec
+    old = spec.REGION_BITS
+    spec.REGION_BITS = bits
+    try:
+        order, starts = binned_permutation(ids, rows)
+    finally:
+        spec.REGION_BITS = old
+    assert sorted(order.tolist()) == list(range(n))
+    assert starts[0] == 0 and starts[-1] == n
+    assert (starts[1:] >= starts[:-1]).all()
+    for b in range(starts.numel() - 1):
+        sl = order[starts[b]:starts[b + 1]]
+        if sl.numel():
+            assert ((ids[sl] >> bits) == b).all()
+
+
+@settings(max_examples=40, deadline=None)
+@given(sizes=st.lists(st.integers(1, 5000), min_size=1, max_size=40),
+       cap_kb=st.integers(1, 64))
+def test_hvd_bucketize_respects_cap_and_order(sizes, cap_kb):
+    import torch
+
+    from tf_yarn_amd.parallel.hvd import _bucketize
+
+    tensors = [torch.empty(s) for s in sizes]
+    groups = _bucketize(tensors, cap_kb * 1024)
+    flat = [t for g in groups for t in g]
+    assert len(flat) == len(tensors)
+    assert all(a is b for a, b in zip(flat, tensors))  # order preserved
+    for g in groups:
+        nb = sum(t.numel() * t.element_size() for t in g)
+        # a single oversized tensor may exceed the cap alone
+        assert nb <= cap_kb * 1024 or len(g) == 1
+
+
+@settings(max_examples=30, deadline=None)
+@given(world=st.integers(1, 16), F=st.integers(1, 40), b=st.integers(1, 8))
+def test_alltoall_split_arithmetic_consistent(world, F, b):
+    """The sharded-embedding split sizes must agree between sender and
+    receiver for ANY (world, features) combination, even when some
+    ranks own zero features (the ragged dp8 case generalized)."""
+    feats_of = [[f for f in range(F) if f % world == s]
+                for s in range(world)]
+    for me in range(world):
+        f_own = len(feats_of[me])
+        # what each peer s sends me (their batch x my features):
+        sent_to_me = [b * f_own for _ in range(world)]
+        # what I expect from every peer:
+        out_splits = [b * f_own] * world
+        assert sent_to_me == out_splits
+        # what I send to each peer covers my whole batch x their features
+        in_splits = [b * len(feats_of[s]) for s in range(world)]
+        assert sum(in_splits) == b * F
