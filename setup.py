@@ -38,11 +38,13 @@ ext = CUDAExtension(
         os.path.join(CSRC, "wgrad128.hip"),
         os.path.join(CSRC, "wgrad256.hip"),
         os.path.join(CSRC, "gemm_bt.hip"),
+        os.path.join(CSRC, "lt_linear.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3"],
         "nvcc": ["-O3", "--offload-arch=gfx950"],
     },
+    libraries=["hipblaslt"],
 )
 
 setup(

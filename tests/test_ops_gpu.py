@@ -765,3 +765,52 @@ def test_wgrad_out_bf16_matches_fp32_cast():
     bf = C.wgrad_nt256(dy, x, 8, True)
     assert bf.dtype == torch.bfloat16
     assert torch.allclose(bf.float(), f32.to(torch.bfloat16).float())
+
+
+@requires_gpu
+@pytest.mark.parametrize("relu,with_bias", [(True, True), (False, True),
+                                            (False, False)])
+def test_lt_linear_matches_reference(relu, with_bias):
+    import tf_yarn_amd.ops._C as C
+    torch.manual_seed(11)
+    M, K, N = 4096, 432, 1024
+    x = torch.randn(M, K, device="cuda").to(torch.bfloat16)
+    w = (torch.randn(N, K, device="cuda") * 0.05).to(torch.bfloat16)
+    b = torch.randn(N, device="cuda").to(torch.bfloat16) if with_bias \
+        else None
+    y = C.lt_linear(x, w, b, relu)
+    ref = x.float() @ w.float().t()
+    if with_bias:
+        ref = ref + b.float()
+    if relu:
+        ref = torch.relu(ref)
+    rel = ((y.float() - ref).abs().max()
+           / ref.abs().max().clamp_min(1e-6)).item()
+    assert rel < 5e-2, rel
+
+
+@requires_gpu
+def test_linear_bias_relu_lt_path_grads_match_kernel_path():
+    """Forward via the fused hipBLASLt epilogue must produce the same
+    gradients as the matmul+epilogue-kernel path."""
+    import os
+    torch.manual_seed(12)
+    M, K, N = 2048, 512, 256
+    results = {}
+    for flag in ("1", "0"):
+        os.environ["MIYARN_LT_FWD"] = flag
+        x = torch.randn(M, K, device="cuda").to(torch.bfloat16)
+        x.requires_grad_(True)
+        torch.manual_seed(13)
+        w = torch.nn.Parameter(
+            (torch.randn(N, K, device="cuda") * 0.05).to(torch.bfloat16))
+        b = torch.nn.Parameter(
+            torch.randn(N, device="cuda").to(torch.bfloat16))
+        y = ops.linear_bias_relu(x, w, b)
+        y.float().pow(2).mean().backward()
+        results[flag] = (y.detach().float(), x.grad.float(),
+                         w.grad.float(), b.grad.float())
+    os.environ.pop("MIYARN_LT_FWD", None)
+    for a, c in zip(results["1"], results["0"]):
+        assert torch.allclose(a, c, atol=5e-2, rtol=5e-2), \
+            (a - c).abs().max().item()

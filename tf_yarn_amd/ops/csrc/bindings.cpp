@@ -68,6 +68,10 @@ std::vector<torch::Tensor> bias_relu_bwd_db(torch::Tensor dy,
                                             torch::Tensor y,
                                             bool dbias_bf16);
 torch::Tensor reduce_splitk(torch::Tensor part, bool out_bf16);
+
+// lt_linear.hip
+torch::Tensor lt_linear(torch::Tensor x, torch::Tensor w,
+                        c10::optional<torch::Tensor> bias, bool relu);
 torch::Tensor col_reduce_dot(torch::Tensor x, torch::Tensor dy);
 torch::Tensor row_dot(torch::Tensor x, torch::Tensor w,
                       c10::optional<torch::Tensor> bias);
@@ -110,6 +114,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Sparse embedding grad scatter into dense grad table");
   m.def("bias_relu_fwd", &bias_relu_fwd, "Fused bias+ReLU forward");
   m.def("bias_relu_bwd", &bias_relu_bwd, "Fused ReLU backward");
+  m.def("lt_linear", &lt_linear,
+        "hipBLASLt linear with fused bias(+ReLU) epilogue",
+        py::arg("x"), py::arg("w"), py::arg("bias") = py::none(),
+        py::arg("relu") = false);
   m.def("reduce_splitk", &reduce_splitk,
         "Split-K partial reduction with fused output cast",
         py::arg("part"), py::arg("out_bf16") = false);
