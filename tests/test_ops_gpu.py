@@ -794,11 +794,11 @@ def test_linear_bias_relu_lt_path_grads_match_kernel_path():
     """Forward via the fused hipBLASLt epilogue must produce the same
     gradients as the matmul+epilogue-kernel path."""
     import os
-    torch.manual_seed(12)
     M, K, N = 2048, 512, 256
     results = {}
     for flag in ("1", "0"):
         os.environ["MIYARN_LT_FWD"] = flag
+        torch.manual_seed(12)
         x = torch.randn(M, K, device="cuda").to(torch.bfloat16)
         x.requires_grad_(True)
         torch.manual_seed(13)
