@@ -39,6 +39,10 @@ def _check_json_line(stdout: str, n_gpus: int):
     assert result["config"]["parallelism"] == f"dp{n_gpus}"
     assert result["config"]["global_batch"] == \
         result["config"]["per_gpu_batch"] * n_gpus
+    # round-2 diagnostics for the unattended scale run
+    assert "alltoall_mode" in result["config"]
+    assert "rank_spread_ms_per_step" in result["config"]
+    assert "comm_breakdown" in result
     return result
 
 
