@@ -34,9 +34,15 @@ INIT_SEED = 20240913
 def _worker_fn(client, cluster_tasks, experiment, task_type: str,
                rank: int, world_size: int, device: str) -> None:
     """Reference ``gloo_allred_task.py:36-92``."""
+    # RCCL needs a DISTINCT GPU per rank (same rule as the reference's
+    # worker backend pick, ``pytorch/tasks/worker.py:171-174``): when
+    # ranks share a device (e.g. a 1-GPU box running a multi-task
+    # topology), fall back to gloo — compute stays on the GPU, only the
+    # collectives stage through CPU.
+    use_rccl = (device.startswith("cuda")
+                and world_size <= torch.cuda.device_count())
     comm.init_process_group(rank=rank, world_size=world_size,
-                            backend="nccl" if device.startswith("cuda")
-                            else "gloo",
+                            backend="nccl" if use_rccl else "gloo",
                             device=device, kv_client=client,
                             group_name="allred")
     try:

@@ -52,10 +52,16 @@ def _get_device(gpu_ids: List[int], local_rank: int) -> str:
     return "cpu"
 
 
-def _get_collective_ops_backend(device: str) -> str:
-    """nccl (=RCCL) for GPU processes, gloo otherwise
-    (reference ``worker.py:171-174``)."""
-    return "nccl" if device.startswith("cuda") else "gloo"
+def _get_collective_ops_backend(device: str,
+                                world_size: int = 1) -> str:
+    """nccl (=RCCL) only when every process can own a distinct GPU,
+    gloo otherwise — the reference's procs-vs-GPUs rule
+    (``worker.py:171-174``); RCCL refuses duplicate devices in one
+    communicator, so ranks sharing a GPU must stage collectives
+    through gloo (compute stays on the GPU)."""
+    if device.startswith("cuda") and             world_size <= torch.cuda.device_count():
+        return "nccl"
+    return "gloo"
 
 
 def _create_dataloader(dataset,
@@ -97,7 +103,7 @@ def _train(client: KVClient,
            rank: int,
            world_size: int) -> None:
     """Per-process training body (reference ``worker.py:94-121``)."""
-    backend = _get_collective_ops_backend(device)
+    backend = _get_collective_ops_backend(device, world_size)
     _task_commons.choose_master(client, rank)
     if device.startswith("cuda"):
         torch.cuda.set_device(torch.device(device))
