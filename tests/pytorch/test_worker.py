@@ -27,7 +27,14 @@ def test_get_device_cpu_when_no_cuda():
 
 
 def test_backend_selection():
-    assert worker._get_collective_ops_backend("cuda:0") == "nccl"
+    import torch
+    n = torch.cuda.device_count()
+    # nccl iff every rank can own a distinct GPU (reference
+    # worker.py:171-174 procs-vs-GPUs rule)
+    assert worker._get_collective_ops_backend(
+        "cuda:0", world_size=max(1, n)) == ("nccl" if n else "gloo")
+    assert worker._get_collective_ops_backend(
+        "cuda:0", world_size=n + 1) == "gloo"
     assert worker._get_collective_ops_backend("cpu") == "gloo"
 
 
