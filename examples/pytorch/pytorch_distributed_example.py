@@ -27,12 +27,17 @@ def train_fn(task_params):
 
     os.environ["MASTER_ADDR"] = task_params.master_address
     os.environ["MASTER_PORT"] = str(task_params.master_port)
+    # RCCL needs a distinct GPU per rank; on a box with fewer GPUs
+    # than ranks, collectives stage through gloo (compute stays on GPU)
     backend = "nccl" if (task_params.gpu_id is not None
-                         and torch.cuda.is_available()) else "gloo"
+                         and torch.cuda.is_available()
+                         and task_params.world_size
+                         <= torch.cuda.device_count()) else "gloo"
     dist.init_process_group(backend, rank=task_params.rank,
                             world_size=task_params.world_size)
     device = (f"cuda:{task_params.gpu_id}"
-              if backend == "nccl" else "cpu")
+              if task_params.gpu_id is not None
+              and torch.cuda.is_available() else "cpu")
     torch.manual_seed(0)
     model = nn.Sequential(nn.Linear(16, 32), nn.ReLU(),
                           nn.Linear(32, 10)).to(device)
