@@ -814,3 +814,28 @@ def test_linear_bias_relu_lt_path_grads_match_kernel_path():
     for a, c in zip(results["1"], results["0"]):
         assert torch.allclose(a, c, atol=5e-2, rtol=5e-2), \
             (a - c).abs().max().item()
+
+
+@requires_gpu
+def test_scatter_auto_skew_selection():
+    """Unset MIYARN_BINNED_SCATTER = auto: uniform ids keep the atomic
+    path, heavily duplicated ids flip to the binned-dedup path — probed
+    once per module from the first update batch."""
+    import os
+    from tf_yarn_amd.models.sharded_embedding import \
+        ShardedCriteoEmbeddings
+    os.environ.pop("MIYARN_BINNED_SCATTER", None)
+
+    def run_one(ids):
+        torch.manual_seed(21)
+        emb = ShardedCriteoEmbeddings([50_000] * 4, 16).cuda()
+        buf = torch.zeros(ids.shape[0], 4 * 16, device="cuda")
+        out, wide = emb(ids, buf, 0)
+        (out.float().pow(2).mean() + wide.float().pow(2).mean()).backward()
+        emb.apply_sparse_updates(0.1)
+        return emb._binned_auto
+
+    uniform = torch.randint(0, 50_000, (8192, 4), device="cuda")
+    assert run_one(uniform) is False
+    skewed = torch.randint(0, 64, (8192, 4), device="cuda")  # dup >> 2
+    assert run_one(skewed) is True

@@ -211,3 +211,15 @@ def test_bench_master_port_rank0_picks_free_port(tmp_path, monkeypatch):
     finally:
         blocker.close()
         __import__("os").environ.pop("MASTER_PORT", None)
+
+
+def test_binned_mode_parsing(monkeypatch):
+    from tf_yarn_amd.models import sharded_embedding as se
+    monkeypatch.setenv("MIYARN_BINNED_SCATTER", "1")
+    assert se._binned_mode() == "on"
+    monkeypatch.setenv("MIYARN_BINNED_SCATTER", "0")
+    assert se._binned_mode() == "off"
+    monkeypatch.delenv("MIYARN_BINNED_SCATTER")
+    assert se._binned_mode() == "auto"
+    monkeypatch.setenv("MIYARN_BINNED_SCATTER", "auto")
+    assert se._binned_mode() == "auto"

@@ -52,10 +52,26 @@ def _world_rank(group) -> Tuple[int, int]:
 _ALLTOALL_MODE: "weakref.WeakKeyDictionary" = weakref.WeakKeyDictionary()
 
 
-def _binned_enabled() -> bool:
+# Binned scatter pays only when ids repeat (the LDS-hash dedup divides
+# the table RMW traffic by the duplication factor but costs a ~200 us
+# permutation).  From the round-2 A/B at the bench shape: atomic 422 us
+# flat; binned perm ~200 + applies that scale with unique/n — break-even
+# near duplication factor ~2.
+SKEW_THRESHOLD = 2.0
+
+
+def _binned_mode() -> str:
+    """MIYARN_BINNED_SCATTER: '1' force on, '0' force off, unset/'auto'
+    = probe the id distribution once and decide (uniform ids keep the
+    measured-fastest atomic kernels; power-law CTR traffic flips to the
+    dedup path automatically)."""
     import os
-    v = os.environ.get("MIYARN_BINNED_SCATTER", "0")
-    return v != "" and v != "0"
+    v = os.environ.get("MIYARN_BINNED_SCATTER", "auto")
+    if v == "1":
+        return "on"
+    if v in ("", "0"):
+        return "off"
+    return "auto"
 
 
 def negotiated_alltoall_mode(group=None) -> str:
@@ -326,6 +342,7 @@ class ShardedCriteoEmbeddings(nn.Module):
         self._last_col_offset = 0
         self._perm_stream = None  # side stream for async pass A
         self._pending_perm = None  # (ids data_ptr, (order, starts), event)
+        self._binned_auto: Optional[bool] = None  # skew probe result
 
     def _ensure_tiled(self, numel: int, device) -> None:
         if self.own_offsets_tiled.numel() < numel:
@@ -380,7 +397,7 @@ class ShardedCriteoEmbeddings(nn.Module):
         ~200 us of latency/atomic-bound work hides under the MFMA-bound
         MLP GEMMs instead of sitting on the backward critical path."""
         if not (flat_ids.is_cuda and self.dim == 16 and ops.HAVE_EXT
-                and torch.is_grad_enabled() and _binned_enabled()):
+                and torch.is_grad_enabled() and self._binned_on(flat_ids)):
             return
         rb = ops.pick_region_bits(self.weight.shape[0], flat_ids.numel())
         if not _perm_stream_enabled():
@@ -400,6 +417,26 @@ class ShardedCriteoEmbeddings(nn.Module):
         ev.record(self._perm_stream)
         self._pending_perm = (flat_ids.data_ptr(), perm, ev)
 
+    def _binned_on(self, flat_ids: torch.Tensor) -> bool:
+        """Resolve the binned-vs-atomic choice (see ``_binned_mode``)."""
+        mode = _binned_mode()
+        if mode == "on":
+            return True
+        if mode == "off":
+            return False
+        if self._binned_auto is None:
+            n = flat_ids.numel()
+            if n == 0:
+                return False
+            uniq = int(torch.unique(flat_ids).numel())
+            dup = n / max(1, uniq)
+            self._binned_auto = dup >= SKEW_THRESHOLD
+            logger.info(
+                "sparse scatter auto-select: duplication factor %.2f -> "
+                "%s path", dup,
+                "binned-dedup" if self._binned_auto else "atomic")
+        return self._binned_auto
+
     def _apply_pending(self, lr: float) -> None:
         scale = 1.0 / self.world
         # Binned scatter (round-2 kernel): one region-binned permutation
@@ -407,8 +444,10 @@ class ShardedCriteoEmbeddings(nn.Module):
         # dedup + exclusive-owner writeback instead of 27M random global
         # atomics.  MIYARN_BINNED_SCATTER=0 restores the atomic kernels
         # for A/B runs.
+        first_ids = self._deep_sink[0][0] if self._deep_sink else None
         use_binned = (self.weight.is_cuda and self.dim == 16
-                      and ops.HAVE_EXT and _binned_enabled())
+                      and ops.HAVE_EXT and first_ids is not None
+                      and self._binned_on(first_ids))
         perms = {}
         if use_binned and self._pending_perm is not None:
             key, perm, ev = self._pending_perm
