@@ -223,3 +223,22 @@ def test_binned_mode_parsing(monkeypatch):
     assert se._binned_mode() == "auto"
     monkeypatch.setenv("MIYARN_BINNED_SCATTER", "auto")
     assert se._binned_mode() == "auto"
+
+
+def test_negotiated_alltoall_mode_states():
+    import torch.distributed as dist
+
+    from tf_yarn_amd.models.sharded_embedding import \
+        negotiated_alltoall_mode
+    # no process group -> unprobed
+    assert negotiated_alltoall_mode() == "unprobed"
+    import os
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT="29561",
+                      RANK="0", WORLD_SIZE="1")
+    dist.init_process_group("gloo")
+    try:
+        assert negotiated_alltoall_mode() == "gloo-emulate"
+    finally:
+        dist.destroy_process_group()
+        for k in ("MASTER_ADDR", "MASTER_PORT", "RANK", "WORLD_SIZE"):
+            os.environ.pop(k, None)
